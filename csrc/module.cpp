@@ -1,0 +1,766 @@
+// torch_quiver — MI355X-native pybind module.
+//
+// Engine classes re-implementing the reference's native surface
+// (torch-quiver srcs/cpp/src/quiver/{cuda/quiver_sample.cu,
+// cuda/quiver_feature.cu, cuda/quiver_comm.cu, quiver.cpp,
+// torch/module.cpp}) as an idiomatic ROCm design: torch caching allocator
+// for scratch, c10::hip streams, RCCL collectives, hipIpc sharing, and the
+// wave64 kernels in *_kernels.hip.
+#include <torch/extension.h>
+
+#include <ATen/Parallel.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <atomic>
+#include <random>
+#include <unordered_map>
+#include <vector>
+
+#include "qk_common.h"
+
+namespace {
+
+#define QK_CHECK_RCCL(expr)                                                   \
+    do {                                                                      \
+        ncclResult_t _r = (expr);                                             \
+        if (_r != ncclSuccess) {                                              \
+            throw std::runtime_error(std::string("RCCL error: ") +            \
+                                     ncclGetErrorString(_r));                 \
+        }                                                                     \
+    } while (0)
+
+inline hipStream_t current_stream() {
+    return c10::hip::getCurrentHIPStream().stream();
+}
+
+inline int64_t next_pow2(int64_t x) {
+    int64_t p = 64;
+    while (p < x) p <<= 1;
+    return p;
+}
+
+struct DeviceScope {
+    int prev = -1;
+    explicit DeviceScope(int dev) {
+        QK_CHECK_HIP(hipGetDevice(&prev));
+        if (dev != prev) QK_CHECK_HIP(hipSetDevice(dev));
+    }
+    ~DeviceScope() { (void)hipSetDevice(prev); }
+};
+
+// Exclusive scan of an int64 device tensor; returns (prefix, total-on-host).
+std::pair<torch::Tensor, int64_t> exclusive_scan_total(
+    const torch::Tensor& vals, hipStream_t stream) {
+    int64_t n = vals.numel();
+    auto prefix = torch::empty_like(vals);
+    auto temp = torch::empty(
+        {(int64_t)qk::scan_temp_bytes(n)},
+        torch::TensorOptions().dtype(torch::kUInt8).device(vals.device()));
+    auto d_total = torch::empty(
+        {1}, torch::TensorOptions().dtype(torch::kInt64).device(vals.device()));
+    qk::launch_exclusive_scan(stream, temp.data_ptr(), temp.numel(),
+                              vals.data_ptr<int64_t>(),
+                              prefix.data_ptr<int64_t>(), n,
+                              d_total.data_ptr<int64_t>());
+    int64_t h_total = 0;
+    QK_CHECK_HIP(hipMemcpyAsync(&h_total, d_total.data_ptr(), sizeof(int64_t),
+                                hipMemcpyDeviceToHost, stream));
+    QK_CHECK_HIP(hipStreamSynchronize(stream));
+    return {prefix, h_total};
+}
+
+// ---------------------------------------------------------------------------
+// GPU sampler engine (parity: reference TorchQuiver, quiver_sample.cu:77-358)
+// ---------------------------------------------------------------------------
+class GpuSampler {
+  public:
+    GpuSampler(torch::Tensor indptr, torch::Tensor indices, torch::Tensor eid,
+               int device, bool dma_mode)
+        : device_(device), dma_(dma_mode) {
+        TORCH_CHECK(indptr.dtype() == torch::kInt64, "indptr must be int64");
+        TORCH_CHECK(indices.dtype() == torch::kInt64, "indices must be int64");
+        DeviceScope g(device_);
+        auto dev = torch::Device(torch::kCUDA, device_);
+        // indptr is always device-resident: it is read twice per seed and is
+        // small (node_count+1).
+        indptr_ = indptr.contiguous().to(dev);
+        node_count_ = indptr_.numel() - 1;
+        has_eid_ = eid.defined() && eid.numel() == indices.numel();
+        if (dma_) {
+            indices_ = indices.contiguous().to(dev);
+            indices_dptr_ = indices_.data_ptr<int64_t>();
+            if (has_eid_) {
+                eid_ = eid.contiguous().to(dev);
+                eid_dptr_ = eid_.data_ptr<int64_t>();
+            }
+        } else {
+            // ZERO_COPY: pin the host CSR columns and read them from kernels
+            // over PCIe (UVA).  Keeps graphs larger than HBM samplable.
+            indices_ = indices.contiguous();
+            TORCH_CHECK(indices_.device().is_cpu(),
+                        "UVA mode expects CPU indices");
+            indices_dptr_ = (int64_t*)register_host(
+                indices_.data_ptr(), indices_.numel() * sizeof(int64_t));
+            if (has_eid_) {
+                eid_ = eid.contiguous();
+                eid_dptr_ = (int64_t*)register_host(
+                    eid_.data_ptr(), eid_.numel() * sizeof(int64_t));
+            }
+        }
+    }
+
+    ~GpuSampler() {
+        for (void* p : registered_) (void)hipHostUnregister(p);
+    }
+
+    void set_seed(uint64_t seed) { rng_counter_.store(seed); }
+
+    std::tuple<torch::Tensor, torch::Tensor> sample_neighbor(
+        int /*stream_id*/, torch::Tensor seeds, int k) {
+        DeviceScope g(device_);
+        auto stream = current_stream();
+        seeds = seeds.contiguous();
+        TORCH_CHECK(seeds.device().is_cuda(), "seeds must be on GPU");
+        int64_t n = seeds.numel();
+        auto opts =
+            torch::TensorOptions().dtype(torch::kInt64).device(seeds.device());
+        auto counts = torch::empty({n}, opts);
+        qk::launch_capped_degree(stream, indptr_.data_ptr<int64_t>(),
+                                 seeds.data_ptr<int64_t>(), n, k,
+                                 counts.data_ptr<int64_t>());
+        auto [prefix, total] = exclusive_scan_total(counts, stream);
+        auto out = torch::empty({total}, opts);
+        uint64_t rs = rng_counter_.fetch_add(0x9e3779b97f4a7c15ULL);
+        qk::launch_sample(stream, indptr_.data_ptr<int64_t>(), indices_dptr_,
+                          nullptr, seeds.data_ptr<int64_t>(), n, k,
+                          prefix.data_ptr<int64_t>(), out.data_ptr<int64_t>(),
+                          nullptr, rs);
+        return {out, counts};
+    }
+
+    // Returns (frontier, row_idx, col_idx): frontier[0:n_seeds] == seeds.
+    std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> reindex_single(
+        torch::Tensor seeds, torch::Tensor nbrs, torch::Tensor counts) {
+        DeviceScope g(device_);
+        auto stream = current_stream();
+        seeds = seeds.contiguous();
+        nbrs = nbrs.contiguous();
+        counts = counts.contiguous();
+        int64_t n = seeds.numel(), m = nbrs.numel();
+        auto opts =
+            torch::TensorOptions().dtype(torch::kInt64).device(seeds.device());
+        auto i32 = opts.dtype(torch::kInt32);
+
+        int64_t capacity = next_pow2(2 * (n + m) + 64);
+        auto keys = torch::empty({capacity}, opts);
+        auto pos = torch::empty({capacity}, i32);
+        auto local = torch::empty({capacity}, i32);
+        qk::launch_reindex_init(stream, keys.data_ptr<int64_t>(),
+                                pos.data_ptr<int32_t>(), capacity);
+        qk::launch_hash_insert(stream, keys.data_ptr<int64_t>(),
+                               pos.data_ptr<int32_t>(), capacity,
+                               seeds.data_ptr<int64_t>(), n,
+                               nbrs.data_ptr<int64_t>(), m);
+        auto flags = torch::empty({n + m}, opts);
+        qk::launch_mark_first(stream, keys.data_ptr<int64_t>(),
+                              pos.data_ptr<int32_t>(), capacity,
+                              seeds.data_ptr<int64_t>(), n,
+                              nbrs.data_ptr<int64_t>(), m,
+                              flags.data_ptr<int64_t>());
+        auto [scanned, unique_total] = exclusive_scan_total(flags, stream);
+        auto frontier = torch::empty({unique_total}, opts);
+        qk::launch_compact_unique(
+            stream, keys.data_ptr<int64_t>(), local.data_ptr<int32_t>(),
+            pos.data_ptr<int32_t>(), capacity, seeds.data_ptr<int64_t>(), n,
+            nbrs.data_ptr<int64_t>(), m, scanned.data_ptr<int64_t>(),
+            flags.data_ptr<int64_t>(), frontier.data_ptr<int64_t>());
+        auto col_idx = torch::empty({m}, opts);
+        qk::launch_lookup_local(stream, keys.data_ptr<int64_t>(),
+                                local.data_ptr<int32_t>(), capacity,
+                                nbrs.data_ptr<int64_t>(), m,
+                                col_idx.data_ptr<int64_t>());
+        auto [prefix, total] = exclusive_scan_total(counts, stream);
+        TORCH_CHECK(total == m, "reindex: counts do not sum to neighbors");
+        auto row_idx = torch::empty({m}, opts);
+        qk::launch_expand_rows(stream, prefix.data_ptr<int64_t>(),
+                               counts.data_ptr<int64_t>(), n,
+                               row_idx.data_ptr<int64_t>());
+        return {frontier, row_idx, col_idx};
+    }
+
+    void cal_neighbor_prob(int /*stream_id*/, torch::Tensor last_prob,
+                           torch::Tensor cur_prob, int k) {
+        DeviceScope g(device_);
+        TORCH_CHECK(dma_ || indices_dptr_, "sampler not initialised");
+        qk::launch_cal_next(current_stream(), indptr_.data_ptr<int64_t>(),
+                            indices_dptr_, last_prob.data_ptr<float>(),
+                            cur_prob.data_ptr<float>(), node_count_, k);
+    }
+
+    int64_t node_count() const { return node_count_; }
+    int64_t edge_count() const { return indices_.numel(); }
+    int device() const { return device_; }
+
+  private:
+    void* register_host(void* p, size_t bytes) {
+        QK_CHECK_HIP(hipHostRegister(
+            p, bytes, hipHostRegisterMapped | hipHostRegisterPortable));
+        registered_.push_back(p);
+        void* dp = nullptr;
+        QK_CHECK_HIP(hipHostGetDevicePointer(&dp, p, 0));
+        return dp;
+    }
+
+    int device_;
+    bool dma_;
+    bool has_eid_ = false;
+    int64_t node_count_ = 0;
+    torch::Tensor indptr_, indices_, eid_;
+    int64_t* indices_dptr_ = nullptr;
+    int64_t* eid_dptr_ = nullptr;
+    std::vector<void*> registered_;
+    std::atomic<uint64_t> rng_counter_{0x853c49e6748fea9bULL};
+};
+
+// ---------------------------------------------------------------------------
+// CPU sampler engine (parity: reference CPUQuiver, quiver.cpp:11-85)
+// ---------------------------------------------------------------------------
+class CpuSampler {
+  public:
+    CpuSampler(torch::Tensor indptr, torch::Tensor indices)
+        : indptr_(indptr.contiguous()), indices_(indices.contiguous()) {
+        TORCH_CHECK(indptr_.device().is_cpu() && indices_.device().is_cpu(),
+                    "CpuSampler expects CPU tensors");
+        TORCH_CHECK(indptr_.dtype() == torch::kInt64 &&
+                    indices_.dtype() == torch::kInt64);
+        node_count_ = indptr_.numel() - 1;
+    }
+
+    std::tuple<torch::Tensor, torch::Tensor> sample_neighbor(
+        torch::Tensor seeds, int k) {
+        seeds = seeds.contiguous().cpu();
+        int64_t n = seeds.numel();
+        const int64_t* sp = seeds.data_ptr<int64_t>();
+        const int64_t* indptr = indptr_.data_ptr<int64_t>();
+        const int64_t* indices = indices_.data_ptr<int64_t>();
+        auto counts = torch::empty({n}, seeds.options());
+        int64_t* cp = counts.data_ptr<int64_t>();
+        at::parallel_for(0, n, 512, [&](int64_t b, int64_t e) {
+            for (int64_t i = b; i < e; ++i) {
+                int64_t deg = indptr[sp[i] + 1] - indptr[sp[i]];
+                cp[i] = (k >= 0 && deg > k) ? k : deg;
+            }
+        });
+        std::vector<int64_t> prefix(n + 1, 0);
+        for (int64_t i = 0; i < n; ++i) prefix[i + 1] = prefix[i] + cp[i];
+        auto out = torch::empty({prefix[n]}, seeds.options());
+        int64_t* op = out.data_ptr<int64_t>();
+        at::parallel_for(0, n, 64, [&](int64_t b, int64_t e) {
+            thread_local std::mt19937_64 gen(std::random_device{}());
+            std::vector<int64_t> slot;
+            for (int64_t i = b; i < e; ++i) {
+                int64_t v = sp[i], beg = indptr[v];
+                int64_t deg = indptr[v + 1] - beg;
+                int64_t* dst = op + prefix[i];
+                if (deg <= k) {
+                    for (int64_t j = 0; j < deg; ++j) dst[j] = indices[beg + j];
+                } else {
+                    slot.resize(k);
+                    for (int64_t j = 0; j < k; ++j) slot[j] = j;
+                    for (int64_t j = k; j < deg; ++j) {
+                        int64_t r = std::uniform_int_distribution<int64_t>(
+                            0, j)(gen);
+                        if (r < k) slot[r] = j;
+                    }
+                    for (int64_t j = 0; j < k; ++j)
+                        dst[j] = indices[beg + slot[j]];
+                }
+            }
+        });
+        return {out, counts};
+    }
+
+    std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> reindex_single(
+        torch::Tensor seeds, torch::Tensor nbrs, torch::Tensor counts) {
+        seeds = seeds.contiguous().cpu();
+        nbrs = nbrs.contiguous().cpu();
+        counts = counts.contiguous().cpu();
+        int64_t n = seeds.numel(), m = nbrs.numel();
+        const int64_t* sp = seeds.data_ptr<int64_t>();
+        const int64_t* np = nbrs.data_ptr<int64_t>();
+        const int64_t* cp = counts.data_ptr<int64_t>();
+        std::unordered_map<int64_t, int64_t> table;
+        table.reserve(2 * (n + m));
+        std::vector<int64_t> frontier;
+        frontier.reserve(n + m);
+        for (int64_t i = 0; i < n; ++i) {
+            auto [it, fresh] = table.emplace(sp[i], (int64_t)frontier.size());
+            if (fresh) frontier.push_back(sp[i]);
+        }
+        auto col = torch::empty({m}, seeds.options());
+        auto row = torch::empty({m}, seeds.options());
+        int64_t* colp = col.data_ptr<int64_t>();
+        int64_t* rowp = row.data_ptr<int64_t>();
+        for (int64_t j = 0; j < m; ++j) {
+            auto [it, fresh] = table.emplace(np[j], (int64_t)frontier.size());
+            if (fresh) frontier.push_back(np[j]);
+            colp[j] = it->second;
+        }
+        int64_t off = 0;
+        for (int64_t i = 0; i < n; ++i)
+            for (int64_t j = 0; j < cp[i]; ++j) rowp[off++] = i;
+        TORCH_CHECK(off == m, "reindex: counts do not sum to neighbors");
+        auto fr = torch::from_blob(frontier.data(),
+                                   {(int64_t)frontier.size()},
+                                   seeds.options())
+                      .clone();
+        return {fr, row, col};
+    }
+
+    int64_t node_count() const { return node_count_; }
+    int64_t edge_count() const { return indices_.numel(); }
+
+  private:
+    torch::Tensor indptr_, indices_;
+    int64_t node_count_;
+};
+
+// ---------------------------------------------------------------------------
+// ShardTensor (parity: reference quiver_feature.cu:20-473), hipIpc + xGMI P2P
+// ---------------------------------------------------------------------------
+struct ShardItem {
+    void* dptr = nullptr;     // device-visible base pointer
+    int64_t rows = 0;
+    int device = -1;          // -1 == pinned host
+    bool owned_hip = false;   // free with hipFree
+    bool from_ipc = false;    // close with hipIpcCloseMemHandle
+    torch::Tensor keeper;     // keeps host/tensor storage alive
+    void* host_reg = nullptr; // hipHostUnregister on destruction
+};
+
+class ShardTensorItem {
+  public:
+    int device = 0;
+    int64_t rows = 0;
+    std::vector<int64_t> row_shape;
+    torch::Dtype dtype = torch::kFloat32;
+    std::string handle;  // hipIpcMemHandle_t bytes
+
+    py::tuple share_ipc() {
+        return py::make_tuple(device, py::bytes(handle), rows,
+                              row_shape, (int)dtype);
+    }
+    static ShardTensorItem from_ipc(py::tuple t) {
+        ShardTensorItem it;
+        it.device = t[0].cast<int>();
+        it.handle = t[1].cast<std::string>();
+        it.rows = t[2].cast<int64_t>();
+        it.row_shape = t[3].cast<std::vector<int64_t>>();
+        it.dtype = (torch::Dtype)t[4].cast<int>();
+        return it;
+    }
+};
+
+class ShardTensor {
+  public:
+    explicit ShardTensor(int current_device) : device_(current_device) {}
+
+    ~ShardTensor() {
+        for (auto& s : shards_) {
+            if (s.owned_hip) {
+                DeviceScope g(s.device);
+                (void)hipFree(s.dptr);
+            } else if (s.from_ipc) {
+                DeviceScope g(device_);
+                (void)hipIpcCloseMemHandle(s.dptr);
+            }
+            if (s.host_reg) (void)hipHostUnregister(s.host_reg);
+        }
+    }
+
+    void append(torch::Tensor tensor, int target_device) {
+        tensor = tensor.contiguous();
+        TORCH_CHECK(tensor.dim() >= 1, "need at least 1-d tensor");
+        init_row_meta(tensor);
+        ShardItem item;
+        item.rows = tensor.size(0);
+        item.device = target_device;
+        size_t bytes = (size_t)tensor.numel() * tensor.element_size();
+        if (target_device >= 0) {
+            DeviceScope g(target_device);
+            // raw hipMalloc (not the torch pool) so hipIpcGetMemHandle works
+            // on the allocation base for cross-process sharing.
+            QK_CHECK_HIP(hipMalloc(&item.dptr, bytes));
+            if (tensor.device().is_cpu()) {
+                QK_CHECK_HIP(hipMemcpy(item.dptr, tensor.data_ptr(), bytes,
+                                       hipMemcpyHostToDevice));
+            } else {
+                QK_CHECK_HIP(hipMemcpy(item.dptr, tensor.data_ptr(), bytes,
+                                       hipMemcpyDeviceToDevice));
+            }
+            item.owned_hip = true;
+        } else {
+            TORCH_CHECK(tensor.device().is_cpu(),
+                        "host shard expects a CPU tensor");
+            // zero-copy pinned host tier: register the tensor's own memory
+            // (shared-memory friendly) and read it from gather kernels.
+            hipError_t err = hipHostRegister(
+                tensor.data_ptr(), bytes,
+                hipHostRegisterMapped | hipHostRegisterPortable);
+            if (err == hipErrorHostMemoryAlreadyRegistered) {
+                (void)hipGetLastError();
+            } else {
+                QK_CHECK_HIP(err);
+                item.host_reg = tensor.data_ptr();
+            }
+            void* dp = nullptr;
+            QK_CHECK_HIP(hipHostGetDevicePointer(&dp, tensor.data_ptr(), 0));
+            item.dptr = dp;
+            item.keeper = tensor;
+        }
+        shards_.push_back(std::move(item));
+    }
+
+    void append_item(const ShardTensorItem& it) {
+        // open a peer shard exported from another process
+        init_row_meta_from(it.row_shape, it.dtype);
+        ShardItem item;
+        item.rows = it.rows;
+        item.device = it.device;
+        item.from_ipc = true;
+        DeviceScope g(device_);
+        hipIpcMemHandle_t h;
+        TORCH_CHECK(it.handle.size() == sizeof(h), "bad ipc handle");
+        memcpy(&h, it.handle.data(), sizeof(h));
+        QK_CHECK_HIP(hipIpcOpenMemHandle(&item.dptr, h,
+                                         hipIpcMemLazyEnablePeerAccess));
+        shards_.push_back(std::move(item));
+    }
+
+    std::vector<ShardTensorItem> share_ipc() {
+        std::vector<ShardTensorItem> out;
+        for (auto& s : shards_) {
+            if (s.device < 0) continue;  // host tier travels via torch shm
+            TORCH_CHECK(s.owned_hip,
+                        "can only export shards this process allocated");
+            ShardTensorItem it;
+            it.device = s.device;
+            it.rows = s.rows;
+            it.row_shape = row_shape_;
+            it.dtype = dtype_;
+            hipIpcMemHandle_t h;
+            DeviceScope g(s.device);
+            QK_CHECK_HIP(hipIpcGetMemHandle(&h, s.dptr));
+            it.handle.assign((char*)&h, sizeof(h));
+            out.push_back(std::move(it));
+        }
+        return out;
+    }
+
+    torch::Tensor gather(torch::Tensor indices) {
+        return gather_on(device_, indices);
+    }
+
+    // Run the gather with `dev` as the executing GPU (used by the
+    // cross-clique fallback: remote device reads its own HBM, result is
+    // copied back by the python layer).
+    torch::Tensor gather_on(int dev, torch::Tensor indices) {
+        DeviceScope g(dev);
+        indices = indices.contiguous();
+        TORCH_CHECK(indices.device().is_cuda(), "indices must be on GPU");
+        int64_t n = indices.numel();
+        std::vector<int64_t> shape = {n};
+        shape.insert(shape.end(), row_shape_.begin(), row_shape_.end());
+        auto out = torch::empty(
+            shape, torch::TensorOptions().dtype(dtype_).device(
+                       torch::Device(torch::kCUDA, dev)));
+        auto spec = build_spec(dev);
+        auto stream = c10::hip::getCurrentHIPStream(dev).stream();
+        qk::launch_gather(stream, spec, indices.data_ptr<int64_t>(), n,
+                          (char*)out.data_ptr());
+        return out;
+    }
+
+    void scatter_update(torch::Tensor indices, torch::Tensor src) {
+        DeviceScope g(device_);
+        indices = indices.contiguous();
+        src = src.contiguous();
+        auto spec = build_spec(device_);
+        qk::launch_scatter(current_stream(), spec,
+                           indices.data_ptr<int64_t>(), indices.numel(),
+                           (const char*)src.data_ptr());
+    }
+
+    // Which shards are directly readable from `dev` (bit per shard).
+    uint32_t access_mask_on(int dev) const {
+        return build_spec(dev).access_mask;
+    }
+    std::vector<int64_t> shard_ends() const {
+        std::vector<int64_t> ends;
+        int64_t acc = 0;
+        for (auto& s : shards_) {
+            acc += s.rows;
+            ends.push_back(acc);
+        }
+        return ends;
+    }
+
+    std::vector<int64_t> shape() const {
+        std::vector<int64_t> s = {total_rows()};
+        s.insert(s.end(), row_shape_.begin(), row_shape_.end());
+        return s;
+    }
+    int64_t total_rows() const {
+        int64_t t = 0;
+        for (auto& s : shards_) t += s.rows;
+        return t;
+    }
+    int64_t size(int dim) const { return shape().at(dim); }
+    int device() const { return device_; }
+    int shard_count() const { return (int)shards_.size(); }
+    std::vector<int64_t> shard_rows() const {
+        std::vector<int64_t> r;
+        for (auto& s : shards_) r.push_back(s.rows);
+        return r;
+    }
+    std::vector<int> shard_devices() const {
+        std::vector<int> r;
+        for (auto& s : shards_) r.push_back(s.device);
+        return r;
+    }
+
+  private:
+    void init_row_meta(const torch::Tensor& t) {
+        std::vector<int64_t> rs(t.sizes().begin() + 1, t.sizes().end());
+        init_row_meta_from(rs, t.scalar_type());
+    }
+    void init_row_meta_from(const std::vector<int64_t>& rs, torch::Dtype dt) {
+        if (row_shape_init_) {
+            TORCH_CHECK(rs == row_shape_ && dt == dtype_,
+                        "all shards must share row shape and dtype");
+            return;
+        }
+        row_shape_ = rs;
+        dtype_ = dt;
+        row_bytes_ = torch::elementSize(dt);
+        for (auto d : rs) row_bytes_ *= d;
+        row_shape_init_ = true;
+    }
+
+    qk::GatherSpec build_spec(int dev) const {
+        TORCH_CHECK((int)shards_.size() <= qk::kMaxShards, "too many shards");
+        qk::GatherSpec spec{};
+        spec.nshards = (int)shards_.size();
+        spec.row_bytes = row_bytes_;
+        spec.access_mask = 0;
+        int64_t acc = 0;
+        DeviceScope g(dev);
+        for (int i = 0; i < spec.nshards; ++i) {
+            acc += shards_[i].rows;
+            spec.ends[i] = acc;
+            spec.ptrs[i] = (const char*)shards_[i].dptr;
+            bool ok = true;
+            int sd = shards_[i].device;
+            if (sd >= 0 && sd != dev) {
+                int can = 0;
+                QK_CHECK_HIP(hipDeviceCanAccessPeer(&can, dev, sd));
+                if (can) {
+                    // lazy xGMI peer enable (idempotent)
+                    hipError_t err = hipDeviceEnablePeerAccess(sd, 0);
+                    if (err != hipSuccess &&
+                        err != hipErrorPeerAccessAlreadyEnabled)
+                        QK_CHECK_HIP(err);
+                    (void)hipGetLastError();
+                }
+                ok = can || shards_[i].from_ipc;
+            }
+            if (ok) spec.access_mask |= (1u << i);
+        }
+        return spec;
+    }
+
+    int device_;
+    std::vector<ShardItem> shards_;
+    std::vector<int64_t> row_shape_;
+    torch::Dtype dtype_ = torch::kFloat32;
+    int64_t row_bytes_ = 0;
+    bool row_shape_init_ = false;
+};
+
+void init_p2p(const std::vector<int>& devices) {
+    // On an 8x MI355X node every pair is xGMI-connected: enable the full
+    // clique (reference init_p2p, quiver_feature.cu:378-421; no NVLink-style
+    // two-clique hardcode here).
+    for (int a : devices) {
+        DeviceScope g(a);
+        for (int b : devices) {
+            if (a == b) continue;
+            int can = 0;
+            QK_CHECK_HIP(hipDeviceCanAccessPeer(&can, a, b));
+            if (can) {
+                hipError_t err = hipDeviceEnablePeerAccess(b, 0);
+                if (err != hipSuccess &&
+                    err != hipErrorPeerAccessAlreadyEnabled)
+                    QK_CHECK_HIP(err);
+                (void)hipGetLastError();
+            }
+        }
+    }
+}
+
+bool can_device_access_peer(int a, int b) {
+    if (a == b) return true;
+    int can = 0;
+    QK_CHECK_HIP(hipDeviceCanAccessPeer(&can, a, b));
+    return can != 0;
+}
+
+// ---------------------------------------------------------------------------
+// RCCL communicator (parity: reference NcclComm, quiver_comm.cu:9-100)
+// ---------------------------------------------------------------------------
+py::bytes create_nccl_id() {
+    ncclUniqueId id;
+    QK_CHECK_RCCL(ncclGetUniqueId(&id));
+    return py::bytes((const char*)&id, sizeof(id));
+}
+
+class RcclComm {
+  public:
+    RcclComm(int rank, int ws, py::bytes id_bytes) : rank_(rank), ws_(ws) {
+        std::string s = id_bytes;
+        TORCH_CHECK(s.size() == sizeof(ncclUniqueId), "bad rccl id");
+        ncclUniqueId id;
+        memcpy(&id, s.data(), sizeof(id));
+        QK_CHECK_RCCL(ncclCommInitRank(&comm_, ws_, id, rank_));
+    }
+    ~RcclComm() {
+        if (comm_) (void)ncclCommDestroy(comm_);
+    }
+
+    int rank() const { return rank_; }
+    int size() const { return ws_; }
+
+    void send(torch::Tensor t, int dst) {
+        QK_CHECK_RCCL(ncclSend(t.data_ptr(), t.numel(), dtype_of(t), dst,
+                               comm_, current_stream()));
+    }
+    void recv(torch::Tensor t, int src) {
+        QK_CHECK_RCCL(ncclRecv(t.data_ptr(), t.numel(), dtype_of(t), src,
+                               comm_, current_stream()));
+    }
+    void allreduce(torch::Tensor t) {
+        QK_CHECK_RCCL(ncclAllReduce(t.data_ptr(), t.data_ptr(), t.numel(),
+                                    dtype_of(t), ncclSum, comm_,
+                                    current_stream()));
+    }
+    void allgather(torch::Tensor src, torch::Tensor dst) {
+        QK_CHECK_RCCL(ncclAllGather(src.data_ptr(), dst.data_ptr(),
+                                    src.numel(), dtype_of(src), comm_,
+                                    current_stream()));
+    }
+    void alltoall(torch::Tensor src, torch::Tensor dst) {
+        int64_t per = src.numel() / ws_;
+        size_t esz = src.element_size();
+        QK_CHECK_RCCL(ncclGroupStart());
+        for (int r = 0; r < ws_; ++r) {
+            QK_CHECK_RCCL(ncclSend((char*)src.data_ptr() + r * per * esz, per,
+                                   dtype_of(src), r, comm_, current_stream()));
+            QK_CHECK_RCCL(ncclRecv((char*)dst.data_ptr() + r * per * esz, per,
+                                   dtype_of(dst), r, comm_, current_stream()));
+        }
+        QK_CHECK_RCCL(ncclGroupEnd());
+    }
+    void group_start() { QK_CHECK_RCCL(ncclGroupStart()); }
+    void group_end() { QK_CHECK_RCCL(ncclGroupEnd()); }
+
+  private:
+    static ncclDataType_t dtype_of(const torch::Tensor& t) {
+        switch (t.scalar_type()) {
+            case torch::kFloat32: return ncclFloat32;
+            case torch::kFloat16: return ncclFloat16;
+            case torch::kBFloat16: return ncclBfloat16;
+            case torch::kInt64: return ncclInt64;
+            case torch::kInt32: return ncclInt32;
+            case torch::kUInt8: return ncclUint8;
+            default: TORCH_CHECK(false, "unsupported dtype for rccl");
+        }
+    }
+    int rank_, ws_;
+    ncclComm_t comm_ = nullptr;
+};
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.doc() = "MI355X-native GNN sampling / feature-collection engine";
+
+    py::class_<GpuSampler>(m, "Quiver")
+        .def("sample_neighbor", &GpuSampler::sample_neighbor)
+        .def("reindex_single", &GpuSampler::reindex_single)
+        .def("cal_neighbor_prob", &GpuSampler::cal_neighbor_prob)
+        .def("set_seed", &GpuSampler::set_seed)
+        .def("node_count", &GpuSampler::node_count)
+        .def("edge_count", &GpuSampler::edge_count)
+        .def("device", &GpuSampler::device);
+
+    m.def("device_quiver_from_csr_array",
+          [](torch::Tensor indptr, torch::Tensor indices, torch::Tensor eid,
+             int device, bool dma) {
+              return new GpuSampler(indptr, indices, eid, device, dma);
+          },
+          py::return_value_policy::take_ownership);
+
+    py::class_<CpuSampler>(m, "CPUQuiver")
+        .def("sample_neighbor", &CpuSampler::sample_neighbor)
+        .def("reindex_single", &CpuSampler::reindex_single)
+        .def("node_count", &CpuSampler::node_count)
+        .def("edge_count", &CpuSampler::edge_count);
+
+    m.def("cpu_quiver_from_csr_array",
+          [](torch::Tensor indptr, torch::Tensor indices) {
+              return new CpuSampler(indptr, indices);
+          },
+          py::return_value_policy::take_ownership);
+
+    py::class_<ShardTensorItem>(m, "ShardTensorItem")
+        .def(py::init<>())
+        .def("share_ipc", &ShardTensorItem::share_ipc)
+        .def_static("from_ipc", &ShardTensorItem::from_ipc)
+        .def_readwrite("device", &ShardTensorItem::device);
+
+    py::class_<ShardTensor>(m, "ShardTensor")
+        .def(py::init<int>())
+        .def("append", &ShardTensor::append)
+        .def("append_item", &ShardTensor::append_item)
+        .def("share_ipc", &ShardTensor::share_ipc)
+        .def("__getitem__", &ShardTensor::gather)
+        .def("gather", &ShardTensor::gather)
+        .def("gather_on", &ShardTensor::gather_on)
+        .def("access_mask_on", &ShardTensor::access_mask_on)
+        .def("shard_ends", &ShardTensor::shard_ends)
+        .def("scatter_update", &ShardTensor::scatter_update)
+        .def("shape", &ShardTensor::shape)
+        .def("size", &ShardTensor::size)
+        .def("device", &ShardTensor::device)
+        .def("shard_count", &ShardTensor::shard_count)
+        .def("shard_rows", &ShardTensor::shard_rows)
+        .def("shard_devices", &ShardTensor::shard_devices);
+
+    m.def("init_p2p", &init_p2p);
+    m.def("can_device_access_peer", &can_device_access_peer);
+
+    m.def("create_nccl_id", &create_nccl_id);
+    py::class_<RcclComm>(m, "NcclComm")
+        .def(py::init<int, int, py::bytes>())
+        .def("rank", &RcclComm::rank)
+        .def("size", &RcclComm::size)
+        .def("send", &RcclComm::send)
+        .def("recv", &RcclComm::recv)
+        .def("allreduce", &RcclComm::allreduce)
+        .def("allgather", &RcclComm::allgather)
+        .def("alltoall", &RcclComm::alltoall)
+        .def("group_start", &RcclComm::group_start)
+        .def("group_end", &RcclComm::group_end);
+}

@@ -1,0 +1,116 @@
+// Common declarations shared by the HIP kernel TUs and the torch-binding TU.
+//
+// Design: kernel TUs (sample_kernels.hip, reindex_kernels.hip,
+// gather_kernels.hip) include only <hip/hip_runtime.h> and expose C-ABI
+// launchers taking raw pointers + a hipStream_t.  module.cpp owns all
+// torch::Tensor plumbing and memory allocation (torch caching allocator).
+//
+// Capability parity map (reference: quiver-team/torch-quiver):
+//   sample_kernels.hip  ~ srcs/cpp/include/quiver/cuda_random.cu.hpp +
+//                         quiver.cu.hpp new_sample path (re-designed wave64)
+//   reindex_kernels.hip ~ srcs/cpp/include/quiver/reindex.cu.hpp
+//   gather_kernels.hip  ~ srcs/cpp/include/quiver/shard_tensor.cu.hpp
+#pragma once
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include <stdexcept>
+#include <string>
+
+#define QK_CHECK_HIP(expr)                                                     \
+    do {                                                                       \
+        hipError_t _e = (expr);                                                \
+        if (_e != hipSuccess) {                                                \
+            throw std::runtime_error(std::string("HIP error at " __FILE__ ":") \
+                                     + std::to_string(__LINE__) + ": "         \
+                                     + hipGetErrorString(_e));                 \
+        }                                                                      \
+    } while (0)
+
+namespace qk {
+
+constexpr int kWaveSize = 64;  // CDNA4 wavefront
+
+// ---------- sampling (sample_kernels.hip) ----------
+
+// counts[i] = degree(seeds[i]); capped[i] = min(counts[i], k). k<0 => no cap.
+void launch_capped_degree(hipStream_t s, const int64_t* indptr,
+                          const int64_t* seeds, int64_t n, int k,
+                          int64_t* capped);
+
+// Exclusive scan over n int64 values; also writes total = sum to d_total.
+// temp_bytes(n) tells the caller how much scratch to allocate.
+size_t scan_temp_bytes(int64_t n);
+void launch_exclusive_scan(hipStream_t s, void* temp, size_t temp_bytes,
+                           const int64_t* in, int64_t* out, int64_t n,
+                           int64_t* d_total);
+
+// Wavefront-subgroup neighbor sampling (reservoir, without replacement).
+// out_nbrs[prefix[i] .. prefix[i]+capped[i]) = sampled neighbor ids of seeds[i].
+// If out_eids != nullptr also emits the matching edge ids (eid_base==nullptr
+// means eid == CSR position).
+void launch_sample(hipStream_t s, const int64_t* indptr, const int64_t* indices,
+                   const int64_t* eid_base, const int64_t* seeds, int64_t n,
+                   int k, const int64_t* prefix, int64_t* out_nbrs,
+                   int64_t* out_eids, uint64_t rng_seed);
+
+// Access-probability propagation (one hop):
+// cur[v] = 1 - (1 - last[v]) * prod_{u in N(v)} (1 - last[u]*min(1, k/deg(u)))
+void launch_cal_next(hipStream_t s, const int64_t* indptr,
+                     const int64_t* indices, const float* last, float* cur,
+                     int64_t node_count, int k);
+
+// ---------- reindex (reindex_kernels.hip) ----------
+
+// Open-addressing hash table working set; all buffers caller-allocated.
+//   capacity: power of two >= 2*(n_seeds+n_nbrs)
+//   keys:   int64[capacity]  (pre-filled with -1 by launch_reindex_init)
+//   pos:    int32[capacity]
+//   local:  int32[capacity]
+//   flags/scan: int64[n_seeds+n_nbrs]
+void launch_reindex_init(hipStream_t s, int64_t* keys, int32_t* pos,
+                         int64_t capacity);
+void launch_hash_insert(hipStream_t s, int64_t* keys, int32_t* pos,
+                        int64_t capacity, const int64_t* seeds, int64_t n_seeds,
+                        const int64_t* nbrs, int64_t n_nbrs);
+void launch_mark_first(hipStream_t s, const int64_t* keys, const int32_t* pos,
+                       int64_t capacity, const int64_t* seeds, int64_t n_seeds,
+                       const int64_t* nbrs, int64_t n_nbrs, int64_t* flags);
+void launch_compact_unique(hipStream_t s, const int64_t* keys, int32_t* local,
+                           const int32_t* pos, int64_t capacity,
+                           const int64_t* seeds, int64_t n_seeds,
+                           const int64_t* nbrs, int64_t n_nbrs,
+                           const int64_t* scanned_flags, const int64_t* flags,
+                           int64_t* out_nodes);
+void launch_lookup_local(hipStream_t s, const int64_t* keys,
+                         const int32_t* local, int64_t capacity,
+                         const int64_t* nbrs, int64_t n_nbrs, int64_t* col_idx);
+// row_idx[prefix[i]+j] = i  for j < counts[i]
+void launch_expand_rows(hipStream_t s, const int64_t* prefix,
+                        const int64_t* counts, int64_t n_seeds,
+                        int64_t* row_idx);
+
+// ---------- feature gather (gather_kernels.hip) ----------
+
+constexpr int kMaxShards = 16;
+
+// One virtual row-major tensor made of up to kMaxShards row ranges living in
+// local HBM, peer-GPU HBM (xGMI) or pinned host memory (zero-copy).
+struct GatherSpec {
+    const char* ptrs[kMaxShards];  // base pointer of shard s (device-visible)
+    int64_t ends[kMaxShards];      // exclusive prefix of row counts
+    uint32_t access_mask;          // bit s set => shard s readable from here
+    int nshards;
+    int64_t row_bytes;
+};
+
+// out[i] = row indices[i] of the virtual tensor; rows whose shard is not
+// accessible are left untouched (python layer fills them via a peer pass).
+void launch_gather(hipStream_t s, const GatherSpec& spec,
+                   const int64_t* indices, int64_t n, char* out);
+
+// Scatter-style update used by the python layer for cache fill / tests:
+// shard-resident rows only.  dst row indices[i] <- src[i].
+void launch_scatter(hipStream_t s, const GatherSpec& spec,
+                    const int64_t* indices, int64_t n, const char* src);
+
+}  // namespace qk

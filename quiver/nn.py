@@ -1,0 +1,175 @@
+"""Minimal PyG-compatible GNN layers/models in plain torch.
+
+The reference relies on torch_geometric for its models; this environment
+has no PyG, so the SAGE/GAT model families used by examples, serving and
+bench.py live here.  The layers follow PyG's bipartite convention: forward
+((x_src, x_dst), edge_index, size) with edge_index[0] indexing x_src and
+edge_index[1] indexing x_dst — exactly what GraphSageSampler's adjs feed.
+Aggregations use index_add / scatter_reduce, which lower to native ROCm
+kernels.
+"""
+from typing import Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+__all__ = ["SAGEConv", "GATConv", "GraphSAGE", "GAT"]
+
+
+def _as_pair(x):
+    if isinstance(x, (tuple, list)):
+        return x[0], x[1]
+    return x, x
+
+
+class SAGEConv(nn.Module):
+    """GraphSAGE-mean convolution."""
+
+    def __init__(self, in_channels, out_channels, bias=True):
+        super().__init__()
+        self.lin_l = nn.Linear(in_channels, out_channels, bias=bias)  # neigh
+        self.lin_r = nn.Linear(in_channels, out_channels, bias=False)  # self
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        nn.init.xavier_uniform_(self.lin_l.weight)
+        nn.init.xavier_uniform_(self.lin_r.weight)
+        if self.lin_l.bias is not None:
+            nn.init.zeros_(self.lin_l.bias)
+
+    def forward(self, x, edge_index, size: Tuple[int, int] = None):
+        x_src, x_dst = _as_pair(x)
+        src, dst = edge_index[0], edge_index[1]
+        n_dst = x_dst.size(0) if size is None else int(size[1])
+        # aggregate-then-project: index_add on in_channels, one GEMM after
+        agg = torch.zeros((n_dst, x_src.size(1)), dtype=x_src.dtype,
+                          device=x_src.device)
+        agg.index_add_(0, dst, x_src[src])
+        deg = torch.zeros(n_dst, dtype=x_src.dtype, device=x_src.device)
+        deg.index_add_(0, dst, torch.ones_like(dst, dtype=x_src.dtype))
+        agg = agg / deg.clamp_(min=1).unsqueeze(-1)
+        return self.lin_l(agg) + self.lin_r(x_dst)
+
+
+class GATConv(nn.Module):
+    """Graph attention convolution (multi-head, concat)."""
+
+    def __init__(self, in_channels, out_channels, heads=1, concat=True,
+                 negative_slope=0.2, dropout=0.0, bias=True):
+        super().__init__()
+        self.heads = heads
+        self.out_channels = out_channels
+        self.concat = concat
+        self.negative_slope = negative_slope
+        self.dropout = dropout
+        self.lin = nn.Linear(in_channels, heads * out_channels, bias=False)
+        self.att_src = nn.Parameter(torch.empty(1, heads, out_channels))
+        self.att_dst = nn.Parameter(torch.empty(1, heads, out_channels))
+        out_dim = heads * out_channels if concat else out_channels
+        self.bias = nn.Parameter(torch.zeros(out_dim)) if bias else None
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        nn.init.xavier_uniform_(self.lin.weight)
+        nn.init.xavier_uniform_(self.att_src)
+        nn.init.xavier_uniform_(self.att_dst)
+        if self.bias is not None:
+            nn.init.zeros_(self.bias)
+
+    def forward(self, x, edge_index, size: Tuple[int, int] = None):
+        x_src, x_dst = _as_pair(x)
+        src, dst = edge_index[0], edge_index[1]
+        H, C = self.heads, self.out_channels
+        n_dst = x_dst.size(0) if size is None else int(size[1])
+
+        h_src = self.lin(x_src).view(-1, H, C)
+        h_dst = self.lin(x_dst).view(-1, H, C)
+        alpha_src = (h_src * self.att_src).sum(-1)  # [N_src, H]
+        alpha_dst = (h_dst * self.att_dst).sum(-1)  # [N_dst, H]
+        alpha = alpha_src[src] + alpha_dst[dst]     # [E, H]
+        alpha = F.leaky_relu(alpha, self.negative_slope)
+        # segment softmax over incoming edges of each dst
+        alpha_max = torch.full((n_dst, H), float("-inf"), dtype=alpha.dtype,
+                               device=alpha.device)
+        alpha_max = alpha_max.scatter_reduce(0, dst.unsqueeze(-1).expand_as(
+            alpha), alpha, reduce="amax", include_self=True)
+        alpha = (alpha - alpha_max[dst]).exp()
+        denom = torch.zeros((n_dst, H), dtype=alpha.dtype,
+                            device=alpha.device)
+        denom.index_add_(0, dst, alpha)
+        alpha = alpha / denom[dst].clamp(min=1e-16)
+        if self.training and self.dropout > 0:
+            alpha = F.dropout(alpha, p=self.dropout)
+
+        msg = h_src[src] * alpha.unsqueeze(-1)      # [E, H, C]
+        out = torch.zeros((n_dst, H, C), dtype=msg.dtype, device=msg.device)
+        out.index_add_(0, dst, msg)
+        out = out.reshape(n_dst, H * C) if self.concat else out.mean(1)
+        if self.bias is not None:
+            out = out + self.bias
+        return out
+
+
+class GraphSAGE(nn.Module):
+    """Multi-layer GraphSAGE matching the reference examples' training/
+    inference loop (forward(x, adjs) over per-hop bipartite graphs)."""
+
+    def __init__(self, in_channels, hidden_channels, out_channels,
+                 num_layers=3, dropout=0.5):
+        super().__init__()
+        self.num_layers = num_layers
+        self.dropout = dropout
+        self.convs = nn.ModuleList()
+        if num_layers == 1:
+            self.convs.append(SAGEConv(in_channels, out_channels))
+        else:
+            self.convs.append(SAGEConv(in_channels, hidden_channels))
+            for _ in range(num_layers - 2):
+                self.convs.append(SAGEConv(hidden_channels, hidden_channels))
+            self.convs.append(SAGEConv(hidden_channels, out_channels))
+
+    def forward(self, x, adjs):
+        for i, (edge_index, _, size) in enumerate(adjs):
+            x_target = x[:size[1]]
+            x = self.convs[i]((x, x_target), edge_index, size)
+            if i != self.num_layers - 1:
+                x = F.relu(x)
+                x = F.dropout(x, p=self.dropout, training=self.training)
+        return torch.log_softmax(x, dim=-1)
+
+    def full_forward(self, x, edge_index):
+        n = x.size(0)
+        for i, conv in enumerate(self.convs):
+            x = conv((x, x), edge_index, (n, n))
+            if i != self.num_layers - 1:
+                x = F.relu(x)
+        return torch.log_softmax(x, dim=-1)
+
+
+class GAT(nn.Module):
+    def __init__(self, in_channels, hidden_channels, out_channels,
+                 num_layers=2, heads=4, dropout=0.5):
+        super().__init__()
+        self.num_layers = num_layers
+        self.dropout = dropout
+        self.convs = nn.ModuleList()
+        if num_layers == 1:
+            self.convs.append(GATConv(in_channels, out_channels, heads=1))
+        else:
+            self.convs.append(GATConv(in_channels, hidden_channels,
+                                      heads=heads))
+            for _ in range(num_layers - 2):
+                self.convs.append(GATConv(hidden_channels * heads,
+                                          hidden_channels, heads=heads))
+            self.convs.append(GATConv(hidden_channels * heads, out_channels,
+                                      heads=1))
+
+    def forward(self, x, adjs):
+        for i, (edge_index, _, size) in enumerate(adjs):
+            x_target = x[:size[1]]
+            x = self.convs[i]((x, x_target), edge_index, size)
+            if i != self.num_layers - 1:
+                x = F.elu(x)
+                x = F.dropout(x, p=self.dropout, training=self.training)
+        return torch.log_softmax(x, dim=-1)

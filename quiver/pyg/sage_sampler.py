@@ -1,0 +1,290 @@
+"""PyG-compatible k-hop neighbor samplers.
+
+Capability parity with reference quiver/pyg/sage_sampler.py:
+  - GraphSageSampler: GPU (DMA) / UVA (zero-copy host CSR) / CPU modes,
+    returns (n_id, batch_size, adjs[::-1]) with PyG's Adj convention.
+  - MixedGraphSageSampler + SampleJob: CPU/GPU work-stealing for one epoch.
+The native engines are the wave64 HIP kernels (csrc/sample_kernels.hip,
+csrc/reindex_kernels.hip) and the at::parallel_for CPU engine.
+"""
+import time
+from dataclasses import dataclass
+from typing import Generic, NamedTuple, Tuple, TypeVar
+
+import torch
+import torch.multiprocessing as mp
+
+from .. import _ext
+from ..utils import CSRTopo
+
+T_co = TypeVar("T_co", covariant=True)
+T = TypeVar("T")
+
+__all__ = ["GraphSageSampler", "MixedGraphSageSampler", "SampleJob", "Adj"]
+
+
+class Adj(NamedTuple):
+    edge_index: torch.Tensor
+    e_id: torch.Tensor
+    size: Tuple[int, int]
+
+    def to(self, *args, **kwargs):
+        return Adj(self.edge_index.to(*args, **kwargs),
+                   self.e_id.to(*args, **kwargs), self.size)
+
+
+@dataclass(frozen=True)
+class _FakeDevice(object):
+    pass
+
+
+@dataclass(frozen=True)
+class _StopWork(object):
+    pass
+
+
+class GraphSageSampler:
+    """K-hop CSR neighbor sampler.
+
+    Args:
+        csr_topo: quiver.CSRTopo of the graph to sample from.
+        sizes: fanout per hop, e.g. [15, 10, 5].
+        device: GPU id (ignored in CPU mode).
+        mode: "UVA" (graph pinned in host DRAM, zero-copy sampled from the
+            GPU), "GPU" (graph resident in HBM), or "CPU".
+    """
+
+    def __init__(self, csr_topo: CSRTopo, sizes, device=0, mode="UVA"):
+        assert mode in ["UVA", "GPU", "CPU"], f"invalid mode: {mode}"
+        self.csr_topo = csr_topo
+        self.sizes = list(sizes)
+        self.mode = mode
+        self.quiver = None
+        self.device = device
+        self.ipc_handle_ = None
+
+    def lazy_init_quiver(self):
+        if self.quiver is not None:
+            return
+        if self.mode == "CPU":
+            self.device = "cpu"
+            self.quiver = _ext.cpu_quiver_from_csr_array(
+                self.csr_topo.indptr, self.csr_topo.indices)
+        else:
+            self.device = torch.cuda.current_device()
+            eid = torch.zeros(0, dtype=torch.long)
+            self.quiver = _ext.device_quiver_from_csr_array(
+                self.csr_topo.indptr, self.csr_topo.indices, eid, self.device,
+                self.mode != "UVA")
+
+    def sample_layer(self, batch, size):
+        self.lazy_init_quiver()
+        if not isinstance(batch, torch.Tensor):
+            batch = torch.tensor(batch, dtype=torch.long)
+        n_id = batch.to(self.device, non_blocking=False)
+        size = size if size != -1 else self.csr_topo.node_count
+        if self.mode in ("GPU", "UVA"):
+            out, cnt = self.quiver.sample_neighbor(0, n_id, size)
+        else:
+            out, cnt = self.quiver.sample_neighbor(n_id, size)
+        return out, cnt
+
+    def reindex(self, inputs, outputs, counts):
+        return self.quiver.reindex_single(inputs, outputs, counts)
+
+    def sample(self, input_nodes):
+        """Sample a k-hop computational graph rooted at `input_nodes`.
+
+        Returns (n_id, batch_size, adjs) exactly as PyG's NeighborSampler:
+        edge_index[0] = source local id in n_id, edge_index[1] = target.
+        """
+        self.lazy_init_quiver()
+        if not isinstance(input_nodes, torch.Tensor):
+            input_nodes = torch.tensor(input_nodes, dtype=torch.long)
+        nodes = input_nodes.to(self.device)
+        adjs = []
+        batch_size = len(nodes)
+        for size in self.sizes:
+            out, cnt = self.sample_layer(nodes, size)
+            frontier, row_idx, col_idx = self.reindex(nodes, out, cnt)
+            row_idx, col_idx = col_idx, row_idx
+            edge_index = torch.stack([row_idx, col_idx], dim=0)
+            adj_size = torch.LongTensor([frontier.size(0), nodes.size(0)])
+            e_id = torch.tensor([])
+            adjs.append(Adj(edge_index, e_id, adj_size))
+            nodes = frontier
+        return nodes, batch_size, adjs[::-1]
+
+    def sample_prob(self, train_idx, total_node_count):
+        """Multi-hop access probability of every node when seeding from
+        train_idx — drives access-probability feature placement."""
+        self.lazy_init_quiver()
+        if self.mode == "CPU":
+            raise RuntimeError("sample_prob needs a GPU sampler")
+        last_prob = torch.zeros(total_node_count, device=self.device)
+        last_prob[train_idx] = 1
+        for size in self.sizes:
+            cur_prob = torch.zeros(total_node_count, device=self.device)
+            self.quiver.cal_neighbor_prob(0, last_prob, cur_prob, size)
+            last_prob = cur_prob
+        return last_prob
+
+    def share_ipc(self):
+        return self.csr_topo, self.sizes, self.mode
+
+    @classmethod
+    def lazy_from_ipc_handle(cls, ipc_handle):
+        csr_topo, sizes, mode = ipc_handle
+        return cls(csr_topo, sizes, _FakeDevice, mode)
+
+
+class SampleJob(Generic[T_co]):
+    """Abstract job list for MixedGraphSageSampler."""
+
+    def __getitem__(self, index) -> T_co:
+        raise NotImplementedError
+
+    def __len__(self) -> int:
+        raise NotImplementedError
+
+    def shuffle(self) -> None:
+        raise NotImplementedError
+
+
+def cpu_sampler_worker_loop(rank, quiver_sampler, task_queue, result_queue):
+    while True:
+        task = task_queue.get()
+        if isinstance(task, _StopWork):
+            result_queue.put(_StopWork())
+            break
+        res = quiver_sampler.sample(task)
+        result_queue.put(res)
+
+
+class MixedGraphSageSampler:
+    """CPU+GPU work-stealing sampler over a SampleJob for one epoch.
+
+    Modes: UVA_CPU_MIXED / GPU_CPU_MIXED (adaptive split between the device
+    sampler and a pool of CPU sampler processes) and UVA_ONLY / GPU_ONLY.
+    Mirrors reference sage_sampler.py:207-376.
+    """
+
+    def __init__(self, sample_job: SampleJob, num_workers, csr_topo: CSRTopo,
+                 sizes, device=0, mode="UVA_CPU_MIXED"):
+        assert mode in ["UVA_CPU_MIXED", "GPU_CPU_MIXED", "UVA_ONLY",
+                        "GPU_ONLY"], f"invalid mode: {mode}"
+        self.sample_job = sample_job
+        self.num_workers = num_workers
+        self.csr_topo = csr_topo
+        self.sizes = list(sizes)
+        self.mode = mode
+        self.device = device
+        self.device_sampler = None
+        self.cpu_sampler = None
+        self.workers = []
+        self.task_queues = []
+        self.result_queue = None
+        self.inited = False
+
+    def lazy_init(self):
+        if self.inited:
+            return
+        dev_mode = "UVA" if self.mode.startswith("UVA") else "GPU"
+        self.device_sampler = GraphSageSampler(self.csr_topo, self.sizes,
+                                               self.device, dev_mode)
+        if self.mode.endswith("MIXED"):
+            self.cpu_sampler = GraphSageSampler(self.csr_topo, self.sizes,
+                                                mode="CPU")
+            ctx = mp.get_context("spawn")
+            self.result_queue = ctx.Queue()
+            for w in range(self.num_workers):
+                q = ctx.Queue()
+                p = ctx.Process(target=cpu_sampler_worker_loop,
+                                args=(w, self.cpu_sampler, q,
+                                      self.result_queue),
+                                daemon=True)
+                p.start()
+                self.task_queues.append(q)
+                self.workers.append(p)
+        self.inited = True
+
+    def decide_task_num(self, dev_time, dev_tasks, cpu_time, cpu_tasks):
+        # adaptive split: how many tasks CPU workers get per GPU task batch
+        if cpu_time <= 0 or cpu_tasks == 0:
+            return max(1, self.num_workers)
+        dev_rate = dev_tasks / max(dev_time, 1e-9)
+        cpu_rate = cpu_tasks / max(cpu_time, 1e-9)
+        return max(0, int(cpu_rate / max(dev_rate, 1e-9) * dev_tasks / 2))
+
+    def __iter__(self):
+        self.lazy_init()
+        self.sample_job.shuffle()
+        return self.iter_sampler()
+
+    def iter_sampler(self):
+        n = len(self.sample_job)
+        next_task = 0
+        pending_cpu = 0
+        dev_time, dev_done = 1e-9, 0
+        cpu_time, cpu_done = 0.0, 0
+        cpu_t0 = None
+        try:
+            while next_task < n or pending_cpu > 0:
+                # hand a slice to CPU workers
+                if self.task_queues and next_task < n:
+                    want = self.decide_task_num(dev_time, max(dev_done, 1),
+                                                cpu_time, cpu_done)
+                    for w in range(min(want, n - next_task,
+                                       len(self.task_queues))):
+                        self.task_queues[w].put(
+                            self.sample_job[next_task])
+                        next_task += 1
+                        pending_cpu += 1
+                    if cpu_t0 is None and pending_cpu:
+                        cpu_t0 = time.time()
+                # GPU samples inline
+                if next_task < n:
+                    t0 = time.time()
+                    res = self.device_sampler.sample(
+                        self.sample_job[next_task])
+                    dev_time += time.time() - t0
+                    dev_done += 1
+                    next_task += 1
+                    yield res
+                # drain CPU results
+                while pending_cpu > 0:
+                    try:
+                        res = self.result_queue.get(
+                            block=(next_task >= n))
+                    except Exception:
+                        break
+                    if isinstance(res, _StopWork):
+                        continue
+                    pending_cpu -= 1
+                    cpu_done += 1
+                    if cpu_t0 is not None:
+                        cpu_time = time.time() - cpu_t0
+                    yield res
+                    if next_task < n:
+                        break
+        finally:
+            pass
+
+    def shutdown(self):
+        for q in self.task_queues:
+            q.put(_StopWork())
+        for p in self.workers:
+            p.join(timeout=5)
+            if p.is_alive():
+                p.terminate()
+        self.workers = []
+        self.task_queues = []
+        self.inited = False
+
+    def share_ipc(self):
+        return (self.sample_job, self.num_workers, self.csr_topo, self.sizes,
+                self.device, self.mode)
+
+    @classmethod
+    def lazy_from_ipc_handle(cls, ipc_handle):
+        return cls(*ipc_handle)

@@ -1,0 +1,368 @@
+"""GNN serving pipeline: request batching -> hybrid CPU/GPU sampling ->
+inference workers, connected by multiprocessing queues.
+
+Capability parity with reference quiver/serving.py (RequestBatcher,
+HybridSampler, InferenceServer, InferenceServer_Debug).  Differences:
+  - workers stop cleanly on a _Stop sentinel (reference relies on
+    daemon-kill / queue timeouts);
+  - device_list entries may be "cpu" so the whole pipeline is testable on a
+    GPU-less host (model then runs on CPU with the CPU sampler).
+"""
+import os
+import time
+
+import numpy as np
+import torch
+import torch.multiprocessing as mp
+
+from .pyg import GraphSageSampler
+
+__all__ = ["RequestBatcher", "HybridSampler", "InferenceServer",
+           "InferenceServer_Debug"]
+
+
+class _Stop(object):
+    pass
+
+
+def _pin(cpu_range, idx):
+    if cpu_range:
+        try:
+            os.sched_setaffinity(0, [cpu_range[idx % len(cpu_range)]])
+        except OSError:
+            pass
+
+
+class RequestBatcher(object):
+    """Routes incoming node-id request batches to per-device CPU or GPU
+    queues; in Auto mode by predicted sampling work
+    sum(neighbour_num[batch]) vs threshold."""
+
+    def __init__(self, device_num, stream_queue_list, input_proc_per_device,
+                 sample_mode="GPU", request_mode="CPU", threshold=800,
+                 batch_time_limit=10, fixed_batch_size=512,
+                 neighbour_path=None, cpu_range=[]):
+        self.cpu_batched_queue_list = [mp.Manager().Queue()
+                                       for _ in range(device_num)]
+        self.gpu_batched_queue_list = [mp.Manager().Queue()
+                                       for _ in range(device_num)]
+        self.stream_queue_list = stream_queue_list
+        self.sample_mode = sample_mode
+        self.device_num = device_num
+        self.batch_time_limit = batch_time_limit / 1000
+        self.threshold = threshold
+        self.fixed_batch_size = fixed_batch_size
+        self.neighbour_path = neighbour_path
+        self.request_mode = request_mode
+        self.cpu_range = cpu_range
+        self.procs = []
+        target = (self.auto_despatch if sample_mode == "Auto"
+                  else self.fixed_despatch)
+        for i in range(input_proc_per_device * device_num):
+            p = mp.Process(target=target, args=(i,), daemon=True)
+            p.start()
+            self.procs.append(p)
+
+    def fixed_despatch(self, idx):
+        _pin(self.cpu_range, idx)
+        stream_queue = self.stream_queue_list[idx]
+        if self.sample_mode == "CPU":
+            batched_queue = self.cpu_batched_queue_list[idx % self.device_num]
+        else:
+            batched_queue = self.gpu_batched_queue_list[idx % self.device_num]
+        while True:
+            item = stream_queue.get()
+            if isinstance(item, _Stop):
+                batched_queue.put(item)
+                break
+            batched_queue.put(item)
+
+    def auto_despatch(self, idx):
+        _pin(self.cpu_range, idx)
+        stream_queue = self.stream_queue_list[idx]
+        gpu_q = self.gpu_batched_queue_list[idx % self.device_num]
+        cpu_q = self.cpu_batched_queue_list[idx % self.device_num]
+        neighbour_num = np.load(self.neighbour_path)
+        while True:
+            item = stream_queue.get()
+            if isinstance(item, _Stop):
+                gpu_q.put(item)
+                cpu_q.put(item)
+                break
+            if self.request_mode == "Preparation":
+                gpu_q.put(item)
+                cpu_q.put(item)
+                continue
+            predicted_work = np.take(neighbour_num, np.asarray(item)).sum()
+            (gpu_q if predicted_work > self.threshold else cpu_q).put(item)
+
+    def batched_request_queue_list(self):
+        return [self.cpu_batched_queue_list, self.gpu_batched_queue_list]
+
+    def stop(self):
+        for q in self.stream_queue_list:
+            q.put(_Stop())
+
+
+class HybridSampler(object):
+    """Pool of CPU sampler processes feeding per-device sampled queues."""
+
+    def __init__(self, csr_topo, sizes, device_num, worker_num_per_device,
+                 batched_queue_list, cpu_range=[]):
+        self.csr_topo = csr_topo
+        self.csr_topo.share_memory_()
+        self.cpu_range = cpu_range
+        self.device_num = device_num
+        self.cpu_num_workers = device_num * worker_num_per_device
+        self.sizes = sizes
+        self.cpu_batched_queue_list = batched_queue_list[0]
+        self.gpu_batched_queue_list = batched_queue_list[1]
+        self.cpu_sampled_queue_list = [mp.Manager().Queue()
+                                       for _ in range(device_num)]
+        self.procs = []
+
+    def start(self):
+        for i in range(self.cpu_num_workers):
+            p = mp.Process(target=self.cpu_sampler_worker_loop,
+                           args=(i, self.cpu_batched_queue_list,
+                                 self.cpu_sampled_queue_list,
+                                 self.device_num, self.sizes, self.csr_topo),
+                           daemon=True)
+            p.start()
+            self.procs.append(p)
+
+    def cpu_sampler_worker_loop(self, rank, sample_task_queue_list,
+                                result_queue_list, device_num, sizes,
+                                csr_topo):
+        _pin(self.cpu_range, rank)
+        cpu_sampler = GraphSageSampler(csr_topo, sizes, device="cpu",
+                                       mode="CPU")
+        task_queue = sample_task_queue_list[rank % device_num]
+        result_queue = result_queue_list[rank % device_num]
+        while True:
+            start = time.perf_counter()
+            item = task_queue.get()
+            if isinstance(item, _Stop):
+                task_queue.put(item)  # let sibling workers see it too
+                result_queue.put(item)
+                break
+            res = cpu_sampler.sample(item)
+            result_queue.put((res, time.perf_counter() - start))
+
+    def sampled_request_queue_list(self):
+        return [self.cpu_sampled_queue_list, self.gpu_batched_queue_list]
+
+
+def _load_model(model_path, device):
+    if isinstance(model_path, torch.nn.Module):
+        return model_path.to(device)
+    return torch.load(model_path, weights_only=False).to(device)
+
+
+def _resolve_device(device_list, rank):
+    d = device_list[rank % len(device_list)]
+    if d == "cpu":
+        return "cpu", None
+    return f"cuda:{d}", d
+
+
+class InferenceServer(object):
+    """Per-GPU inference workers: GPU workers sample+infer, CPU-fed workers
+    infer batches pre-sampled by HybridSampler."""
+
+    def __init__(self, model_path, device_list, x_feature, task_queue_list,
+                 sample_mode, csr_topo, sizes, ignord_length=100,
+                 proc_num_per_device=0, uva_gpu="GPU"):
+        self.cpu_sampled_queue_list = task_queue_list[0]
+        self.model_path = model_path
+        self.device_list = device_list
+        self.x_feature = x_feature
+        self.gpu_task_queue_list = task_queue_list[1]
+        self.sample_mode = sample_mode
+        self.csr_topo = csr_topo
+        self.sizes = sizes
+        self.ignord_length = ignord_length
+        self.proc_num_per_device = proc_num_per_device
+        self.uva_gpu = uva_gpu
+        self.num_proc = len(self.device_list) * self.proc_num_per_device
+        self.output_queue_list = [mp.Manager().Queue()
+                                  for _ in range(self.num_proc)]
+
+    def start(self, join=True):
+        mp.spawn(self.run,
+                 args=(self.device_list, self.cpu_sampled_queue_list,
+                       self.model_path, self.x_feature,
+                       self.gpu_task_queue_list, self.sample_mode,
+                       self.csr_topo, self.sizes, self.num_proc, self.uva_gpu,
+                       self.output_queue_list),
+                 nprocs=self.num_proc, join=join)
+
+    def run(self, rank, device_list, cpu_sampled_queue_list, model_path,
+            feature, gpu_sample_task_queue_list, sample_mode, csr_topo, sizes,
+            num_proc, uva_gpu, output_queue_list):
+        output_queue = output_queue_list[rank]
+        if sample_mode == "Auto":
+            if rank < num_proc // 2:
+                self.gpu_sampler_inference_loop(
+                    rank, device_list, feature, gpu_sample_task_queue_list,
+                    model_path, csr_topo, sizes, uva_gpu, output_queue)
+            else:
+                self.cpu_sampler_inference_loop(
+                    rank - num_proc // 2, device_list, feature,
+                    cpu_sampled_queue_list, model_path, output_queue)
+        elif sample_mode == "GPU":
+            self.gpu_sampler_inference_loop(
+                rank, device_list, feature, gpu_sample_task_queue_list,
+                model_path, csr_topo, sizes, uva_gpu, output_queue)
+        else:
+            self.cpu_sampler_inference_loop(rank, device_list, feature,
+                                            cpu_sampled_queue_list,
+                                            model_path, output_queue)
+
+    def gpu_sampler_inference_loop(self, rank, device_list, feature,
+                                   gpu_sample_task_queue_list, model_path,
+                                   csr_topo, sizes, sample_mode,
+                                   output_queue):
+        device, dev_id = _resolve_device(device_list, rank)
+        q = gpu_sample_task_queue_list[rank % len(device_list)]
+        mode = "CPU" if device == "cpu" else sample_mode
+        sampler = GraphSageSampler(csr_topo, sizes,
+                                   device=(dev_id if dev_id is not None
+                                           else "cpu"), mode=mode)
+        model = _load_model(model_path, device)
+        model.eval()
+        with torch.no_grad():
+            while True:
+                item = q.get()
+                if isinstance(item, _Stop):
+                    q.put(item)
+                    output_queue.put(item)
+                    break
+                sample_task = torch.as_tensor(item)
+                n_id, batch_size, adjs = sampler.sample(sample_task)
+                adjs = [adj.to(device) for adj in adjs]
+                x_input = feature[n_id].to(device)
+                out = model(x_input, adjs)
+                output_queue.put(out.cpu())
+
+    def cpu_sampler_inference_loop(self, rank, device_list, feature,
+                                   cpu_sampled_queue_list, model_path,
+                                   output_queue):
+        device, _ = _resolve_device(device_list, rank)
+        q = cpu_sampled_queue_list[rank % len(device_list)]
+        model = _load_model(model_path, device)
+        model.eval()
+        with torch.no_grad():
+            while True:
+                item = q.get()
+                if isinstance(item, _Stop):
+                    q.put(item)
+                    output_queue.put(item)
+                    break
+                n_id, batch_size, adjs = item[0]
+                adjs = [adj.to(device) for adj in adjs]
+                x_input = feature[n_id].to(device)
+                out = model(x_input, adjs)
+                output_queue.put(out.cpu())
+
+    def result_queue_list(self):
+        return self.output_queue_list
+
+
+class InferenceServer_Debug(InferenceServer):
+    """InferenceServer variant with avg/p99 latency + throughput accounting
+    (reference serving.py:236-360)."""
+
+    def __init__(self, model_path, device_list, x_feature, task_queue_list,
+                 sample_mode, csr_topo, sizes, ignord_length=100,
+                 result_path=None, exp_id=0, proc_num_per_device=0,
+                 uva_gpu="GPU"):
+        super().__init__(model_path, device_list, x_feature, task_queue_list,
+                         sample_mode, csr_topo, sizes, ignord_length,
+                         proc_num_per_device, uva_gpu)
+        self.result_path = result_path
+        self.exp_id = exp_id
+
+    @staticmethod
+    def _report(result, ignord_length, tag, rank, res_path, exp_id):
+        if not result:
+            return None
+        arr = np.array(result[ignord_length:] if
+                       len(result) > ignord_length else result)
+        lat = arr[:, 2] - arr[:, 0]
+        avg_latency = np.average(lat, axis=0, weights=arr[:, 3])
+        tp99_latency = np.percentile(lat, 99, axis=0) * 1000
+        span = max(np.max(arr[:, 2]) - np.min(arr[:, 2]), 1e-9)
+        throughput = np.sum(arr[:, 3]) / span
+        total = np.sum(arr[:, 3])
+        print(f"{tag} Rank {rank}: Avg Latency: {avg_latency}, "
+              f"TP99 Latency: {tp99_latency} ms, Throughput: {throughput}, "
+              f"Total: {total}", flush=True)
+        if res_path is not None:
+            np.save(os.path.join(res_path, f"{exp_id}_{tag}_{rank}"), arr)
+        return dict(avg_latency=float(avg_latency),
+                    tp99_latency_ms=float(tp99_latency),
+                    throughput=float(throughput), total=float(total))
+
+    def gpu_sampler_inference_loop(self, rank, device_list, feature,
+                                   gpu_sample_task_queue_list, model_path,
+                                   csr_topo, sizes, sample_mode,
+                                   output_queue):
+        device, dev_id = _resolve_device(device_list, rank)
+        q = gpu_sample_task_queue_list[rank % len(device_list)]
+        mode = "CPU" if device == "cpu" else sample_mode
+        sampler = GraphSageSampler(csr_topo, sizes,
+                                   device=(dev_id if dev_id is not None
+                                           else "cpu"), mode=mode)
+        model = _load_model(model_path, device)
+        model.eval()
+        result = []
+        with torch.no_grad():
+            while True:
+                item = q.get()
+                if isinstance(item, _Stop):
+                    q.put(item)
+                    stats = self._report(result, self.ignord_length, "GPU",
+                                         rank, self.result_path, self.exp_id)
+                    output_queue.put(stats if stats is not None else item)
+                    break
+                start_time = time.perf_counter()
+                sample_task = torch.as_tensor(item)
+                n_id, batch_size, adjs = sampler.sample(sample_task)
+                sample_time = time.perf_counter()
+                adjs = [adj.to(device) for adj in adjs]
+                x_input = feature[n_id].to(device)
+                out = model(x_input, adjs)
+                if device != "cpu":
+                    torch.cuda.synchronize()
+                end_time = time.perf_counter()
+                result.append([start_time, sample_time, end_time, batch_size,
+                               n_id.shape[0]])
+
+    def cpu_sampler_inference_loop(self, rank, device_list, feature,
+                                   cpu_sampled_queue_list, model_path,
+                                   output_queue):
+        device, _ = _resolve_device(device_list, rank)
+        q = cpu_sampled_queue_list[rank % len(device_list)]
+        model = _load_model(model_path, device)
+        model.eval()
+        result = []
+        with torch.no_grad():
+            while True:
+                item = q.get()
+                if isinstance(item, _Stop):
+                    q.put(item)
+                    stats = self._report(result, self.ignord_length, "CPU",
+                                         rank, self.result_path, self.exp_id)
+                    output_queue.put(stats if stats is not None else item)
+                    break
+                start_time = time.perf_counter()
+                n_id, batch_size, adjs = item[0]
+                adjs = [adj.to(device) for adj in adjs]
+                x_input = feature[n_id].to(device)
+                out = model(x_input, adjs)
+                if device != "cpu":
+                    torch.cuda.synchronize()
+                end_time = time.perf_counter()
+                result.append([start_time, start_time, end_time, batch_size,
+                               n_id.shape[0]])

@@ -1,0 +1,61 @@
+import torch
+
+import quiver
+from quiver.utils import parse_size, get_csr_from_coo
+
+
+def test_csr_from_coo_roundtrip():
+    edge_index = torch.tensor([[0, 0, 1, 2, 2, 3], [1, 2, 0, 0, 3, 2]])
+    indptr, indices, eid = get_csr_from_coo(edge_index)
+    assert indptr.tolist() == [0, 2, 3, 5, 6]
+    # neighbors per node as sets
+    assert sorted(indices[0:2].tolist()) == [1, 2]
+    assert indices[2].item() == 0
+    assert sorted(indices[3:5].tolist()) == [0, 3]
+    assert indices[5].item() == 2
+    # eid maps back to original edge positions
+    src, dst = edge_index
+    for pos in range(6):
+        e = eid[pos].item()
+        row = int(torch.searchsorted(indptr, pos, right=True)) - 1
+        assert src[e] == row and dst[e] == indices[pos]
+
+
+def test_csr_topo_properties():
+    edge_index = torch.tensor([[0, 1, 1, 2], [1, 0, 2, 1]])
+    topo = quiver.CSRTopo(edge_index)
+    assert topo.node_count == 3
+    assert topo.edge_count == 4
+    assert topo.degree.tolist() == [1, 2, 1]
+    topo.share_memory_()
+    assert topo.indptr.is_shared()
+
+
+def test_csr_topo_from_arrays():
+    indptr = torch.tensor([0, 1, 2])
+    indices = torch.tensor([1, 0])
+    topo = quiver.CSRTopo(indptr=indptr, indices=indices)
+    assert topo.node_count == 2
+    assert topo.edge_count == 2
+
+
+def test_parse_size():
+    assert parse_size(1024) == 1024
+    assert parse_size("1K") == 1024
+    assert parse_size("2M") == 2 * 2**20
+    assert parse_size("1.5G") == int(1.5 * 2**30)
+    assert parse_size("200MB") == 200 * 2**20
+    assert parse_size("0") == 0
+
+
+def test_reindex_by_config():
+    edge_index = torch.tensor([[0, 0, 0, 1, 2], [1, 2, 3, 0, 0]])
+    topo = quiver.CSRTopo(edge_index)
+    feat = torch.arange(4, dtype=torch.float32).unsqueeze(1).repeat(1, 3)
+    new_feat, order = quiver.utils.reindex_feature(topo, feat, 0.0)
+    # order[v] = new row of node v; reordering must be a permutation
+    assert sorted(order.tolist()) == [0, 1, 2, 3]
+    for v in range(4):
+        assert torch.equal(new_feat[order[v]], feat[v])
+    # node 0 has the highest degree -> first row after degree sort
+    assert order[0].item() == 0
