@@ -117,8 +117,8 @@ def main():
         model = torch.nn.parallel.DistributedDataParallel(
             model, device_ids=[local_rank])
     opt = torch.optim.Adam(model.parameters(), lr=3e-3)
-    y = torch.randint(0, N_CLASSES, (args.nodes,),
-                      generator=g)
+    # labels resident on GPU: indexing them with n_id stays device-side
+    y = torch.randint(0, N_CLASSES, (args.nodes,), generator=g).to(device)
 
     # per-rank seed stream over the train range (hot head of the id space,
     # like real train_idx after degree reorder)
@@ -135,7 +135,7 @@ def main():
         x = feature[n_id]
         adjs = [adj.to(device) for adj in adjs]
         out = model(x, adjs)
-        loss = F.nll_loss(out, y[n_id[:bs].cpu()].to(device))
+        loss = F.nll_loss(out, y[n_id[:bs]])
         opt.zero_grad(set_to_none=True)
         loss.backward()
         opt.step()

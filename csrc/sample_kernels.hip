@@ -82,7 +82,10 @@ sample_kernel(const int64_t* __restrict__ indptr,
             for (int i = lane; i < k; i += SUB) slot[i] = i;
             // lanes of one subgroup are in one wavefront: LDS writes above are
             // visible to the atomics below without a barrier (lock-step exec).
-            const uint64_t base = rng_seed ^ ((uint64_t)v << 1);
+            // RNG stream is keyed by ROW (not node id): duplicate seeds in a
+            // batch must sample independently.
+            const uint64_t base =
+                rng_seed + (uint64_t)row * 0x9e3779b97f4a7c15ULL;
             for (int64_t j = k + lane; j < deg; j += SUB) {
                 uint64_t h = splitmix64(base + (uint64_t)j * 0x632be59bd9b4e019ULL);
                 int64_t r = bounded_rand(h, j + 1);

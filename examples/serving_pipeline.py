@@ -1,0 +1,82 @@
+#!/usr/bin/env python3
+"""GNN serving (BASELINE config 5): RequestBatcher -> HybridSampler ->
+InferenceServer_Debug with workload-aware Auto routing.
+
+Requests stream node-id batches; Auto mode predicts per-batch sampling work
+from the offline neighbour_num table and routes heavy batches to GPU
+workers, light ones to the CPU sampler pool.  Prints avg/p99 latency and
+throughput per worker at the end.
+"""
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np
+import torch
+import torch.multiprocessing as mp
+
+import quiver
+from quiver.nn import GraphSAGE
+
+
+def main(n=100_000, dim=64, requests=500, batch=64, device_list=None,
+         tmpdir="/tmp/quiver_serving"):
+    os.makedirs(tmpdir, exist_ok=True)
+    use_gpu = torch.cuda.is_available()
+    if device_list is None:
+        device_list = [0, 1] if (use_gpu and torch.cuda.device_count() > 1) \
+            else ([0] if use_gpu else ["cpu"])
+    rng = np.random.default_rng(0)
+    deg = np.maximum((rng.pareto(1.3, n) * 8).astype(np.int64), 1)
+    indptr = np.zeros(n + 1, dtype=np.int64)
+    np.cumsum(deg, out=indptr[1:])
+    indices = rng.integers(0, n, int(indptr[-1]), dtype=np.int64)
+    csr_topo = quiver.CSRTopo(indptr=torch.from_numpy(indptr),
+                              indices=torch.from_numpy(indices))
+    sizes = [10, 5]
+    x = torch.randn(n, dim)
+
+    # offline workload predictor for Auto routing
+    nbr_path = os.path.join(tmpdir, "neighbour_num.npy")
+    quiver.generate_neighbour_num(n, csr_topo, sizes, nbr_path,
+                                  device_list=["cpu"], mode="CPU",
+                                  sample=False)
+    model_path = os.path.join(tmpdir, "model.pt")
+    torch.save(GraphSAGE(dim, 128, 16, num_layers=2), model_path)
+
+    nproc = len(device_list)
+    stream_queues = [mp.Manager().Queue() for _ in range(nproc)]
+    batcher = quiver.RequestBatcher(device_num=nproc,
+                                    stream_queue_list=stream_queues,
+                                    input_proc_per_device=1,
+                                    sample_mode="Auto", threshold=2000,
+                                    neighbour_path=nbr_path)
+    hybrid = quiver.HybridSampler(csr_topo, sizes, device_num=nproc,
+                                  worker_num_per_device=2,
+                                  batched_queue_list=
+                                  batcher.batched_request_queue_list())
+    hybrid.start()
+    server = quiver.InferenceServer_Debug(
+        model_path, device_list, x, hybrid.sampled_request_queue_list(),
+        sample_mode="Auto", csr_topo=csr_topo, sizes=sizes,
+        ignord_length=20, proc_num_per_device=2, uva_gpu="UVA")
+    import threading
+    t = threading.Thread(target=server.start, kwargs=dict(join=True))
+    t.start()
+
+    time.sleep(5)  # workers warm up
+    for i in range(requests):
+        ids = rng.integers(0, n, batch)
+        stream_queues[i % nproc].put(ids)
+        time.sleep(0.002)  # ~500 QPS offered load per queue
+    time.sleep(3)
+    batcher.stop()
+    t.join(timeout=120)
+    print("serving example done")
+
+
+if __name__ == "__main__":
+    # neighbour_num generation needs the sampler; do it inline first
+    main()

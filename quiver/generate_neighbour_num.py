@@ -70,10 +70,13 @@ def generate_neighbour_num(node_num, edge_index, sizes, result_path,
                            device_list=("cpu",), parallel=False, mode="CPU",
                            num_proc=1, reverse=False, sample=False):
     """Compute neighbour_num[v] for all v and save as .npy at result_path."""
-    if reverse:
-        edge_index = edge_index.flip(0) if isinstance(
-            edge_index, torch.Tensor) else edge_index[::-1]
-    csr_topo = CSRTopo(edge_index, node_count=node_num)
+    if isinstance(edge_index, CSRTopo):
+        csr_topo = edge_index
+    else:
+        if reverse:
+            edge_index = edge_index.flip(0) if isinstance(
+                edge_index, torch.Tensor) else edge_index[::-1]
+        csr_topo = CSRTopo(edge_index, node_count=node_num)
     if not parallel or not sample:
         arr = single_generate_neighbour_num(0, node_num, csr_topo, 1, sizes,
                                             mode, result_path, device_list,
