@@ -17,6 +17,7 @@ from .serving import (RequestBatcher, HybridSampler, InferenceServer,
                       InferenceServer_Debug)
 from . import multiprocessing  # noqa: F401  registers ForkingPickler reducers
 from . import nn  # model zoo (SAGE/GAT) — PyG-compatible layers
+from . import trace  # pipeline-stage tracing (QUIVER_TRACE=1)
 
 __version__ = "0.1.0"
 

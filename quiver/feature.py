@@ -13,6 +13,7 @@ import numpy as np
 import torch
 
 from .shard_tensor import ShardTensor, ShardTensorConfig
+from .trace import trace_scope
 from .utils import Topo, CSRTopo, reindex_feature, parse_size
 
 __all__ = ["Feature", "DistFeature", "PartitionInfo", "DeviceConfig"]
@@ -215,9 +216,10 @@ class Feature(object):
         self.lazy_init_from_ipc_handle()
         node_idx = node_idx.to(self.rank)
         if self.mmap_handle_ is None:
-            if self.feature_order is not None:
-                node_idx = self.feature_order[node_idx]
-            return self._shard_tensor()[node_idx]
+            with trace_scope("feature.gather"):
+                if self.feature_order is not None:
+                    node_idx = self.feature_order[node_idx]
+                return self._shard_tensor()[node_idx]
         # disk tier: disk_map < 0 -> mmap row, >= 0 -> in-memory row
         num_nodes = node_idx.size(0)
         disk_index = self.disk_map[node_idx]

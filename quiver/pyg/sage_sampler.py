@@ -15,6 +15,7 @@ import torch
 import torch.multiprocessing as mp
 
 from .. import _ext
+from ..trace import trace_scope
 from ..utils import CSRTopo
 
 T_co = TypeVar("T_co", covariant=True)
@@ -105,8 +106,10 @@ class GraphSageSampler:
         adjs = []
         batch_size = len(nodes)
         for size in self.sizes:
-            out, cnt = self.sample_layer(nodes, size)
-            frontier, row_idx, col_idx = self.reindex(nodes, out, cnt)
+            with trace_scope("sampler.sample_layer"):
+                out, cnt = self.sample_layer(nodes, size)
+            with trace_scope("sampler.reindex"):
+                frontier, row_idx, col_idx = self.reindex(nodes, out, cnt)
             row_idx, col_idx = col_idx, row_idx
             edge_index = torch.stack([row_idx, col_idx], dim=0)
             adj_size = torch.LongTensor([frontier.size(0), nodes.size(0)])
