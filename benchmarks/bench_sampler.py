@@ -29,10 +29,11 @@ SHAPES = {
 }
 
 
-def make_graph(nodes, edges, seed=0):
+def make_graph(nodes, edges, seed=0, max_deg=20_000):
     rng = np.random.default_rng(seed)
     raw = rng.pareto(1.3, nodes) + 0.1
     deg = np.maximum((raw * (edges / raw.sum())).astype(np.int64), 1)
+    np.clip(deg, 1, max_deg, out=deg)
     deg = -np.sort(-deg)
     indptr = np.zeros(nodes + 1, dtype=np.int64)
     np.cumsum(deg, out=indptr[1:])

@@ -13,6 +13,7 @@
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
 
+#include <algorithm>
 #include <atomic>
 #include <random>
 #include <unordered_map>
@@ -266,6 +267,18 @@ class CpuSampler {
                 int64_t* dst = op + prefix[i];
                 if (deg <= k) {
                     for (int64_t j = 0; j < deg; ++j) dst[j] = indices[beg + j];
+                } else if (k <= 256 && deg > 2 * (int64_t)k) {
+                    // Floyd's uniform k-subset: O(k^2), degree-independent
+                    slot.clear();
+                    for (int64_t j = deg - k; j < deg; ++j) {
+                        int64_t t = std::uniform_int_distribution<int64_t>(
+                            0, j)(gen);
+                        bool found = std::find(slot.begin(), slot.end(), t) !=
+                                     slot.end();
+                        slot.push_back(found ? j : t);
+                    }
+                    for (int64_t j = 0; j < k; ++j)
+                        dst[j] = indices[beg + slot[j]];
                 } else {
                     slot.resize(k);
                     for (int64_t j = 0; j < k; ++j) slot[j] = j;

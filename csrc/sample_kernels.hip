@@ -70,6 +70,9 @@ sample_kernel(const int64_t* __restrict__ indptr,
         const int64_t beg = indptr[v];
         const int64_t deg = indptr[v + 1] - beg;
         const int64_t off = prefix[row];
+        // RNG stream is keyed by ROW (not node id): duplicate seeds in a
+        // batch must sample independently.
+        const uint64_t base = rng_seed + (uint64_t)row * 0x9e3779b97f4a7c15ULL;
         if (deg <= (int64_t)k) {
             // copy all neighbors
             for (int64_t j = lane; j < deg; j += SUB) {
@@ -77,15 +80,37 @@ sample_kernel(const int64_t* __restrict__ indptr,
                 if (WITH_EID)
                     out_eids[off + j] = eid_base ? eid_base[beg + j] : beg + j;
             }
+        } else if (k <= 128 && deg > 2 * (int64_t)k) {
+            // Floyd's uniform k-subset: O(k^2) independent of degree — on a
+            // power-law graph hub seeds would otherwise serialize the whole
+            // wave for O(deg) iterations.  One lane draws (sequential
+            // dependency on the chosen set), all lanes gather.
+            int* slot = slots + sub_id * k;
+            if (lane == 0) {
+                int c = 0;
+                for (int64_t j = deg - k; j < deg; ++j) {
+                    uint64_t h = splitmix64(
+                        base + (uint64_t)j * 0x632be59bd9b4e019ULL);
+                    int t = (int)bounded_rand(h, j + 1);
+                    bool found = false;
+                    for (int i = 0; i < c; ++i) {
+                        if (slot[i] == t) { found = true; break; }
+                    }
+                    slot[c++] = found ? (int)j : t;
+                }
+            }
+            // lane 0's LDS writes are visible to its wave without a barrier
+            for (int i = lane; i < k; i += SUB) {
+                int64_t p = beg + slot[i];
+                out_nbrs[off + i] = indices[p];
+                if (WITH_EID) out_eids[off + i] = eid_base ? eid_base[p] : p;
+            }
         } else {
+            // cooperative reservoir with LDS atomicMax replacement order
             int* slot = slots + sub_id * k;
             for (int i = lane; i < k; i += SUB) slot[i] = i;
             // lanes of one subgroup are in one wavefront: LDS writes above are
             // visible to the atomics below without a barrier (lock-step exec).
-            // RNG stream is keyed by ROW (not node id): duplicate seeds in a
-            // batch must sample independently.
-            const uint64_t base =
-                rng_seed + (uint64_t)row * 0x9e3779b97f4a7c15ULL;
             for (int64_t j = k + lane; j < deg; j += SUB) {
                 uint64_t h = splitmix64(base + (uint64_t)j * 0x632be59bd9b4e019ULL);
                 int64_t r = bounded_rand(h, j + 1);
