@@ -83,6 +83,8 @@ def main():
                    choices=["device_replicate", "p2p_clique_replicate"])
     p.add_argument("--nodes", type=int, default=N_NODES)
     p.add_argument("--edges", type=int, default=N_EDGES)
+    p.add_argument("--no-overlap", action="store_true",
+                   help="disable the sample+gather / compute prefetch overlap")
     args = p.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -139,10 +141,7 @@ def main():
         print(f"# setup done in {time.perf_counter()-t0:.1f}s "
               f"(edges={m/1e6:.1f}M)", flush=True)
 
-    def step(seeds):
-        n_id, bs, adjs = sampler.sample(seeds)
-        x = feature[n_id]
-        adjs = [adj.to(device) for adj in adjs]
+    def train_on(n_id, bs, adjs, x):
         out = model(x, adjs)
         loss = F.nll_loss(out, y[n_id[:bs]])
         opt.zero_grad(set_to_none=True)
@@ -150,16 +149,28 @@ def main():
         opt.step()
         return sum(adj.edge_index.shape[1] for adj in adjs)
 
-    for i in range(args.warmup):
-        step(batches[i])
+    def run_range(lo, hi):
+        total = 0
+        if args.no_overlap:
+            for i in range(lo, hi):
+                n_id, bs, adjs = sampler.sample(batches[i])
+                x = feature[n_id]
+                total += train_on(n_id, bs, adjs, x)
+        else:
+            pf = quiver.TrainingPrefetcher(sampler, feature,
+                                           batches[lo:hi], depth=2,
+                                           device=local_rank)
+            for n_id, bs, adjs, x in pf:
+                total += train_on(n_id, bs, adjs, x)
+        return total
+
+    run_range(0, args.warmup)
     if distributed:
         dist.barrier()
     torch.cuda.synchronize()
 
-    edges_done = 0
     t_start = time.perf_counter()
-    for i in range(args.steps):
-        edges_done += step(batches[args.warmup + i])
+    edges_done = run_range(args.warmup, args.warmup + args.steps)
     if distributed:
         dist.barrier()
     torch.cuda.synchronize()
@@ -205,6 +216,7 @@ def main():
                 "sample_mode": args.mode,
                 "cache": args.cache,
                 "cache_policy": args.cache_policy,
+                "overlap": not args.no_overlap,
                 "epoch_seconds_derived": epoch_seconds,
                 "ref_epoch_seconds": ref,
             },

@@ -710,9 +710,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "MI355X-native GNN sampling / feature-collection engine";
 
     py::class_<GpuSampler>(m, "Quiver")
-        .def("sample_neighbor", &GpuSampler::sample_neighbor)
-        .def("reindex_single", &GpuSampler::reindex_single)
-        .def("cal_neighbor_prob", &GpuSampler::cal_neighbor_prob)
+        .def("sample_neighbor", &GpuSampler::sample_neighbor,
+             py::call_guard<py::gil_scoped_release>())
+        .def("reindex_single", &GpuSampler::reindex_single,
+             py::call_guard<py::gil_scoped_release>())
+        .def("cal_neighbor_prob", &GpuSampler::cal_neighbor_prob,
+             py::call_guard<py::gil_scoped_release>())
         .def("set_seed", &GpuSampler::set_seed)
         .def("node_count", &GpuSampler::node_count)
         .def("edge_count", &GpuSampler::edge_count)
@@ -726,8 +729,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::return_value_policy::take_ownership);
 
     py::class_<CpuSampler>(m, "CPUQuiver")
-        .def("sample_neighbor", &CpuSampler::sample_neighbor)
-        .def("reindex_single", &CpuSampler::reindex_single)
+        .def("sample_neighbor", &CpuSampler::sample_neighbor,
+             py::call_guard<py::gil_scoped_release>())
+        .def("reindex_single", &CpuSampler::reindex_single,
+             py::call_guard<py::gil_scoped_release>())
         .def("node_count", &CpuSampler::node_count)
         .def("edge_count", &CpuSampler::edge_count);
 
@@ -748,12 +753,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         .def("append", &ShardTensor::append)
         .def("append_item", &ShardTensor::append_item)
         .def("share_ipc", &ShardTensor::share_ipc)
-        .def("__getitem__", &ShardTensor::gather)
-        .def("gather", &ShardTensor::gather)
-        .def("gather_on", &ShardTensor::gather_on)
+        .def("__getitem__", &ShardTensor::gather,
+             py::call_guard<py::gil_scoped_release>())
+        .def("gather", &ShardTensor::gather,
+             py::call_guard<py::gil_scoped_release>())
+        .def("gather_on", &ShardTensor::gather_on,
+             py::call_guard<py::gil_scoped_release>())
         .def("access_mask_on", &ShardTensor::access_mask_on)
         .def("shard_ends", &ShardTensor::shard_ends)
-        .def("scatter_update", &ShardTensor::scatter_update)
+        .def("scatter_update", &ShardTensor::scatter_update,
+             py::call_guard<py::gil_scoped_release>())
         .def("shape", &ShardTensor::shape)
         .def("size", &ShardTensor::size)
         .def("device", &ShardTensor::device)

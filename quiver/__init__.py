@@ -18,6 +18,7 @@ from .serving import (RequestBatcher, HybridSampler, InferenceServer,
 from . import multiprocessing  # noqa: F401  registers ForkingPickler reducers
 from . import nn  # model zoo (SAGE/GAT) — PyG-compatible layers
 from . import trace  # pipeline-stage tracing (QUIVER_TRACE=1)
+from .loader import TrainingPrefetcher
 
 __version__ = "0.1.0"
 
@@ -28,4 +29,5 @@ __all__ = [
     "p2pCliqueTopo", "init_p2p", "getNcclId", "NcclComm",
     "RequestBatcher", "HybridSampler", "InferenceServer",
     "InferenceServer_Debug", "generate_neighbour_num", "nn",
+    "TrainingPrefetcher",
 ]
