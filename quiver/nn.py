@@ -49,7 +49,10 @@ class SAGEConv(nn.Module):
         deg = torch.zeros(n_dst, dtype=x_src.dtype, device=x_src.device)
         deg.index_add_(0, dst, torch.ones_like(dst, dtype=x_src.dtype))
         agg = agg / deg.clamp_(min=1).unsqueeze(-1)
-        return self.lin_l(agg) + self.lin_r(x_dst)
+        # self path fused into the neighbor GEMM's epilogue via addmm
+        out = self.lin_l(agg)
+        return out.addmm_(x_dst, self.lin_r.weight.t()) if not out.requires_grad \
+            else torch.addmm(out, x_dst, self.lin_r.weight.t())
 
 
 class GATConv(nn.Module):

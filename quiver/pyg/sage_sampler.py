@@ -63,6 +63,10 @@ class GraphSageSampler:
         self.quiver = None
         self.device = device
         self.ipc_handle_ = None
+        # keep the frontier tail ascending: downstream feature gathers and
+        # the next hop's CSR reads then walk memory monotonically (PCIe/L2
+        # locality); seeds stay first, PyG semantics unchanged
+        self.sort_frontier = mode != "CPU"
 
     def lazy_init_quiver(self):
         if self.quiver is not None:
@@ -110,6 +114,18 @@ class GraphSageSampler:
                 out, cnt = self.sample_layer(nodes, size)
             with trace_scope("sampler.reindex"):
                 frontier, row_idx, col_idx = self.reindex(nodes, out, cnt)
+            if self.sort_frontier:
+                bs0 = nodes.size(0)
+                u = frontier.numel()
+                if u > bs0:
+                    tail, perm = torch.sort(frontier[bs0:])
+                    frontier = torch.cat([frontier[:bs0], tail])
+                    inv = torch.empty(u, dtype=torch.long,
+                                      device=frontier.device)
+                    inv[:bs0] = torch.arange(bs0, device=frontier.device)
+                    inv[bs0 + perm] = torch.arange(bs0, u,
+                                                   device=frontier.device)
+                    col_idx = inv[col_idx]
             row_idx, col_idx = col_idx, row_idx
             edge_index = torch.stack([row_idx, col_idx], dim=0)
             adj_size = torch.LongTensor([frontier.size(0), nodes.size(0)])
