@@ -10,6 +10,8 @@
 //  - rows are copied with the widest aligned vector type (16B dwordx4 when
 //    row_bytes % 16 == 0) by a subgroup of lanes sized to the row, so a
 //    wave64 moves up to 1 KiB per instruction; reference copies byte-wise.
+#include <cstdlib>
+
 #include "qk_common.h"
 
 namespace qk {
@@ -54,11 +56,25 @@ copy_rows_kernel(GatherSpec spec, const int64_t* __restrict__ indices,
     }
 }
 
-inline int grid_for(int64_t work, int per_block) {
+inline int grid_for(int64_t work, int per_block, int max_blocks) {
     int64_t blocks = (work + per_block - 1) / per_block;
-    if (blocks > 2048) blocks = 2048;
+    if (blocks > max_blocks) blocks = max_blocks;
     if (blocks < 1) blocks = 1;
     return (int)blocks;
+}
+
+inline int gather_max_blocks(const GatherSpec& spec) {
+    // A gather touching a pinned-host (zero-copy) shard is PCIe-latency
+    // bound: a few hundred blocks keep enough requests in flight to
+    // saturate the link, and capping the grid leaves CUs free so the
+    // prefetch pipeline's gather truly overlaps model kernels on the other
+    // stream.  Pure-HBM/xGMI gathers want the full chip.
+    static int env_cap = [] {
+        const char* e = getenv("QUIVER_GATHER_BLOCKS");
+        return e ? atoi(e) : 0;
+    }();
+    if (env_cap > 0) return env_cap;
+    return spec.has_host_shard ? 640 : 2048;
 }
 
 template <typename VecT, bool SCATTER>
@@ -67,7 +83,7 @@ void dispatch_sub(hipStream_t s, const GatherSpec& spec,
     int64_t nvec = spec.row_bytes / (int64_t)sizeof(VecT);
     int sub = 4;
     while (sub < 64 && sub < nvec) sub *= 2;  // cover the row in ~1 pass
-    int grid = grid_for(n, BLOCK / sub);
+    int grid = grid_for(n, BLOCK / sub, gather_max_blocks(spec));
     switch (sub) {
 #define QK_CASE(S)                                                          \
     case S:                                                                 \

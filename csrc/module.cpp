@@ -568,12 +568,14 @@ class ShardTensor {
         spec.nshards = (int)shards_.size();
         spec.row_bytes = row_bytes_;
         spec.access_mask = 0;
+        spec.has_host_shard = false;
         int64_t acc = 0;
         DeviceScope g(dev);
         for (int i = 0; i < spec.nshards; ++i) {
             acc += shards_[i].rows;
             spec.ends[i] = acc;
             spec.ptrs[i] = (const char*)shards_[i].dptr;
+            if (shards_[i].device < 0) spec.has_host_shard = true;
             bool ok = true;
             int sd = shards_[i].device;
             if (sd >= 0 && sd != dev) {

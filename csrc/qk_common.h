@@ -101,6 +101,7 @@ struct GatherSpec {
     uint32_t access_mask;          // bit s set => shard s readable from here
     int nshards;
     int64_t row_bytes;
+    bool has_host_shard;           // any zero-copy pinned-host tier present
 };
 
 // out[i] = row indices[i] of the virtual tensor; rows whose shard is not
