@@ -21,6 +21,7 @@ KERNEL_SOURCES = [
     "sample_kernels.hip",
     "reindex_kernels.hip",
     "gather_kernels.hip",
+    "segment_kernels.hip",
 ]
 TORCH_SOURCES = ["module.cpp"]
 OUT = os.path.join(ROOT, "torch_quiver.so")

@@ -604,6 +604,41 @@ class ShardTensor {
     bool row_shape_init_ = false;
 };
 
+// ---------------------------------------------------------------------------
+// Fused SAGE-mean aggregation over dst-sorted edges (segment_kernels.hip)
+// ---------------------------------------------------------------------------
+torch::Tensor segment_mean_gather(torch::Tensor x, torch::Tensor src,
+                                  torch::Tensor dst_ptr) {
+    TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kFloat32 && x.dim() == 2);
+    x = x.contiguous();
+    src = src.contiguous();
+    dst_ptr = dst_ptr.contiguous();
+    int64_t n_dst = dst_ptr.numel() - 1;
+    auto out = torch::empty({n_dst, x.size(1)}, x.options());
+    qk::launch_segment_mean_fwd(current_stream(), x.data_ptr<float>(),
+                                src.data_ptr<int64_t>(),
+                                dst_ptr.data_ptr<int64_t>(), n_dst, x.size(1),
+                                out.data_ptr<float>());
+    return out;
+}
+
+torch::Tensor segment_mean_gather_backward(torch::Tensor grad_out,
+                                           torch::Tensor src,
+                                           torch::Tensor dst_ptr,
+                                           int64_t n_src) {
+    grad_out = grad_out.contiguous();
+    src = src.contiguous();
+    dst_ptr = dst_ptr.contiguous();
+    int64_t n_dst = dst_ptr.numel() - 1;
+    auto grad_x = torch::zeros({n_src, grad_out.size(1)}, grad_out.options());
+    qk::launch_segment_mean_bwd(current_stream(),
+                                grad_out.data_ptr<float>(),
+                                src.data_ptr<int64_t>(),
+                                dst_ptr.data_ptr<int64_t>(), n_dst,
+                                grad_out.size(1), grad_x.data_ptr<float>());
+    return grad_x;
+}
+
 void init_p2p(const std::vector<int>& devices) {
     // On an 8x MI355X node every pair is xGMI-connected: enable the full
     // clique (reference init_p2p, quiver_feature.cu:378-421; no NVLink-style
@@ -774,6 +809,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
 
     m.def("init_p2p", &init_p2p);
     m.def("can_device_access_peer", &can_device_access_peer);
+
+    m.def("segment_mean_gather", &segment_mean_gather,
+          py::call_guard<py::gil_scoped_release>());
+    m.def("segment_mean_gather_backward", &segment_mean_gather_backward,
+          py::call_guard<py::gil_scoped_release>());
 
     m.def("create_nccl_id", &create_nccl_id);
     py::class_<RcclComm>(m, "NcclComm")

@@ -114,4 +114,15 @@ void launch_gather(hipStream_t s, const GatherSpec& spec,
 void launch_scatter(hipStream_t s, const GatherSpec& spec,
                     const int64_t* indices, int64_t n, const char* src);
 
+// ---------- fused message-passing aggregation (segment_kernels.hip) ------
+
+// out[d] = mean over edges e in [dst_ptr[d], dst_ptr[d+1]) of x[src[e]]
+void launch_segment_mean_fwd(hipStream_t s, const float* x,
+                             const int64_t* src, const int64_t* dst_ptr,
+                             int64_t n_dst, int64_t dim, float* out);
+// grad_x[src[e]] += grad_out[d] / deg(d)   (grad_x pre-zeroed)
+void launch_segment_mean_bwd(hipStream_t s, const float* grad_out,
+                             const int64_t* src, const int64_t* dst_ptr,
+                             int64_t n_dst, int64_t dim, float* grad_x);
+
 }  // namespace qk

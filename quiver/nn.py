@@ -14,6 +14,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from . import _ext
+
 __all__ = ["SAGEConv", "GATConv", "GraphSAGE", "GAT"]
 
 
@@ -23,11 +25,56 @@ def _as_pair(x):
     return x, x
 
 
-class SAGEConv(nn.Module):
-    """GraphSAGE-mean convolution."""
+class _SegmentMeanAgg(torch.autograd.Function):
+    """Fused mean aggregation over dst-sorted edges (csrc/segment_kernels.hip).
 
-    def __init__(self, in_channels, out_channels, bias=True):
+    Replaces x[src] -> zeros -> index_add -> deg -> div with one kernel in
+    each direction; the layer-1 input features carry no grad, so their
+    backward is skipped entirely.
+    """
+
+    @staticmethod
+    def forward(ctx, x, src, dst_ptr):
+        ctx.save_for_backward(src, dst_ptr)
+        ctx.n_src = x.size(0)
+        return _ext.segment_mean_gather(x, src, dst_ptr)
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        src, dst_ptr = ctx.saved_tensors
+        grad_x = _ext.segment_mean_gather_backward(
+            grad_out.contiguous(), src, dst_ptr, ctx.n_src)
+        return grad_x, None, None
+
+
+def _mean_aggregate(x_src, src, dst, n_dst, sorted_dst=False):
+    """Mean of x_src[src] grouped by dst.  Uses the fused HIP kernel when
+    the caller guarantees dst is sorted ascending (our sampler's layout)
+    on fp32 GPU tensors."""
+    if (sorted_dst and x_src.is_cuda and x_src.dtype == torch.float32
+            and dst.numel() > 0):
+        dst_ptr = torch.searchsorted(
+            dst, torch.arange(n_dst + 1, device=dst.device))
+        return _SegmentMeanAgg.apply(x_src, src, dst_ptr)
+    agg = torch.zeros((n_dst, x_src.size(1)), dtype=x_src.dtype,
+                      device=x_src.device)
+    agg.index_add_(0, dst, x_src[src])
+    deg = torch.zeros(n_dst, dtype=x_src.dtype, device=x_src.device)
+    deg.index_add_(0, dst, torch.ones_like(dst, dtype=x_src.dtype))
+    return agg / deg.clamp_(min=1).unsqueeze(-1)
+
+
+class SAGEConv(nn.Module):
+    """GraphSAGE-mean convolution.
+
+    Set sorted_dst=True when edge_index[1] is sorted ascending (always the
+    case for GraphSageSampler's adjs) to enable the fused segment-mean HIP
+    kernel."""
+
+    def __init__(self, in_channels, out_channels, bias=True,
+                 sorted_dst=False):
         super().__init__()
+        self.sorted_dst = sorted_dst
         self.lin_l = nn.Linear(in_channels, out_channels, bias=bias)  # neigh
         self.lin_r = nn.Linear(in_channels, out_channels, bias=False)  # self
         self.reset_parameters()
@@ -42,13 +89,9 @@ class SAGEConv(nn.Module):
         x_src, x_dst = _as_pair(x)
         src, dst = edge_index[0], edge_index[1]
         n_dst = x_dst.size(0) if size is None else int(size[1])
-        # aggregate-then-project: index_add on in_channels, one GEMM after
-        agg = torch.zeros((n_dst, x_src.size(1)), dtype=x_src.dtype,
-                          device=x_src.device)
-        agg.index_add_(0, dst, x_src[src])
-        deg = torch.zeros(n_dst, dtype=x_src.dtype, device=x_src.device)
-        deg.index_add_(0, dst, torch.ones_like(dst, dtype=x_src.dtype))
-        agg = agg / deg.clamp_(min=1).unsqueeze(-1)
+        # aggregate-then-project: segment mean on in_channels, one GEMM after
+        agg = _mean_aggregate(x_src, src, dst, n_dst,
+                              sorted_dst=self.sorted_dst)
         # self path fused into the neighbor GEMM's epilogue via addmm
         out = self.lin_l(agg)
         return out.addmm_(x_dst, self.lin_r.weight.t()) if not out.requires_grad \
@@ -119,18 +162,23 @@ class GraphSAGE(nn.Module):
     inference loop (forward(x, adjs) over per-hop bipartite graphs)."""
 
     def __init__(self, in_channels, hidden_channels, out_channels,
-                 num_layers=3, dropout=0.5):
+                 num_layers=3, dropout=0.5, sorted_dst=True):
         super().__init__()
         self.num_layers = num_layers
         self.dropout = dropout
         self.convs = nn.ModuleList()
+        # sorted_dst: GraphSageSampler emits dst-sorted adjs -> fused kernel
         if num_layers == 1:
-            self.convs.append(SAGEConv(in_channels, out_channels))
+            self.convs.append(SAGEConv(in_channels, out_channels,
+                                       sorted_dst=sorted_dst))
         else:
-            self.convs.append(SAGEConv(in_channels, hidden_channels))
+            self.convs.append(SAGEConv(in_channels, hidden_channels,
+                                       sorted_dst=sorted_dst))
             for _ in range(num_layers - 2):
-                self.convs.append(SAGEConv(hidden_channels, hidden_channels))
-            self.convs.append(SAGEConv(hidden_channels, out_channels))
+                self.convs.append(SAGEConv(hidden_channels, hidden_channels,
+                                           sorted_dst=sorted_dst))
+            self.convs.append(SAGEConv(hidden_channels, out_channels,
+                                       sorted_dst=sorted_dst))
 
     def forward(self, x, adjs):
         for i, (edge_index, _, size) in enumerate(adjs):
@@ -143,6 +191,9 @@ class GraphSAGE(nn.Module):
 
     def full_forward(self, x, edge_index):
         n = x.size(0)
+        # fused segment path needs dst-sorted edges; sort once up front
+        order = torch.argsort(edge_index[1], stable=True)
+        edge_index = edge_index[:, order]
         for i, conv in enumerate(self.convs):
             x = conv((x, x), edge_index, (n, n))
             if i != self.num_layers - 1:
