@@ -93,6 +93,21 @@ def build(verbose=True):
         run(link)
     print(f"built {OUT}")
 
+    # standalone C++ kernel test binary (no torch); travels with gpurun
+    # snapshots, run by tests/test_gpu_cpp.py on the GPU box
+    cpp_test_src = os.path.join(ROOT, "tests", "cpp", "test_kernels.hip")
+    cpp_test_bin = os.path.join(BUILD, "qk_tests")
+    kernel_objs = [os.path.join(BUILD, s.replace(".hip", ".o"))
+                   for s in KERNEL_SOURCES]
+    if (not os.path.exists(cpp_test_bin)
+            or os.path.getmtime(cpp_test_src) > os.path.getmtime(cpp_test_bin)
+            or any(os.path.getmtime(o) > os.path.getmtime(cpp_test_bin)
+                   for o in kernel_objs)):
+        cpp_test_obj = os.path.join(BUILD, "test_kernels.o")
+        run(common + ["-c", cpp_test_src, "-o", cpp_test_obj])
+        run(common + [cpp_test_obj] + kernel_objs + ["-o", cpp_test_bin])
+    print(f"built {cpp_test_bin}")
+
 
 if __name__ == "__main__":
     build()

@@ -153,6 +153,16 @@ class HybridSampler(object):
         return [self.cpu_sampled_queue_list, self.gpu_batched_queue_list]
 
 
+def _feature_rows(feature, n_id, device):
+    """Index the feature store with a frontier that may live on another
+    device (GPU sampler output vs. plain CPU tensor store and vice versa)."""
+    if isinstance(feature, torch.Tensor):
+        if feature.device != n_id.device:
+            n_id = n_id.to(feature.device)
+        return feature[n_id].to(device)
+    return feature[n_id].to(device)
+
+
 def _load_model(model_path, device):
     if isinstance(model_path, torch.nn.Module):
         return model_path.to(device)
@@ -241,7 +251,7 @@ class InferenceServer(object):
                 sample_task = torch.as_tensor(item)
                 n_id, batch_size, adjs = sampler.sample(sample_task)
                 adjs = [adj.to(device) for adj in adjs]
-                x_input = feature[n_id].to(device)
+                x_input = _feature_rows(feature, n_id, device)
                 out = model(x_input, adjs)
                 output_queue.put(out.cpu())
 
@@ -261,7 +271,7 @@ class InferenceServer(object):
                     break
                 n_id, batch_size, adjs = item[0]
                 adjs = [adj.to(device) for adj in adjs]
-                x_input = feature[n_id].to(device)
+                x_input = _feature_rows(feature, n_id, device)
                 out = model(x_input, adjs)
                 output_queue.put(out.cpu())
 
@@ -331,7 +341,7 @@ class InferenceServer_Debug(InferenceServer):
                 n_id, batch_size, adjs = sampler.sample(sample_task)
                 sample_time = time.perf_counter()
                 adjs = [adj.to(device) for adj in adjs]
-                x_input = feature[n_id].to(device)
+                x_input = _feature_rows(feature, n_id, device)
                 out = model(x_input, adjs)
                 if device != "cpu":
                     torch.cuda.synchronize()
@@ -359,7 +369,7 @@ class InferenceServer_Debug(InferenceServer):
                 start_time = time.perf_counter()
                 n_id, batch_size, adjs = item[0]
                 adjs = [adj.to(device) for adj in adjs]
-                x_input = feature[n_id].to(device)
+                x_input = _feature_rows(feature, n_id, device)
                 out = model(x_input, adjs)
                 if device != "cpu":
                     torch.cuda.synchronize()
