@@ -33,21 +33,24 @@ import torch.nn.functional as F
 import quiver
 from quiver.nn import GraphSAGE
 
-# ogbn-products shape
-N_NODES = 2_449_029
-N_EDGES = 123_718_280
-FEAT_DIM = 100
-N_CLASSES = 47
-N_TRAIN = 196_615
+# graph presets: ogbn-products (default; the reference's headline E2E
+# config) and ogbn-papers100M (reference benchmarks/ogbn-papers100M/
+# dist_sampling_ogb_paper100M_quiver.py:128,192-197 — same fanout/batch,
+# 8G cache, 128 feats; host-DRAM feature spill is the point of the config)
+PRESETS = {
+    "products": dict(nodes=2_449_029, edges=123_718_280, feat_dim=100,
+                     classes=47, train=196_615, cache="196M"),
+    "papers100M": dict(nodes=111_059_956, edges=1_615_685_872, feat_dim=128,
+                       classes=172, train=1_207_179, cache="8G"),
+}
 BATCH = 1024
 FANOUT = [15, 10, 5]
 HIDDEN = 256
-CACHE = "196M"  # ~20% of the 980 MB feature tensor per GPU
 
-REF_EPOCH_SECONDS = {1: 11.1, 2: 5.8, 3: 4.7, 4: 3.25}
+REF_EPOCH_SECONDS = {1: 11.1, 2: 5.8, 3: 4.7, 4: 3.25}  # products only
 
 
-def make_graph(seed=0, nodes=N_NODES, edges=N_EDGES, max_deg=20_000):
+def make_graph(seed=0, nodes=None, edges=None, max_deg=20_000):
     """Directly build a power-law CSR: no COO sort needed.
 
     Degrees are Pareto with the tail clipped at max_deg (ogbn-products'
@@ -78,14 +81,40 @@ def main():
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--batch", type=int, default=BATCH)
     p.add_argument("--mode", default="UVA", choices=["UVA", "GPU"])
-    p.add_argument("--cache", default=CACHE)
+    p.add_argument("--preset", default="products", choices=sorted(PRESETS))
+    p.add_argument("--cache", default=None,
+                   help="per-GPU HBM feature cache (default: preset's)")
     p.add_argument("--cache-policy", default="device_replicate",
                    choices=["device_replicate", "p2p_clique_replicate"])
-    p.add_argument("--nodes", type=int, default=N_NODES)
-    p.add_argument("--edges", type=int, default=N_EDGES)
+    p.add_argument("--nodes", type=int, default=None)
+    p.add_argument("--edges", type=int, default=None)
     p.add_argument("--no-overlap", action="store_true",
                    help="disable the sample+gather / compute prefetch overlap")
     args = p.parse_args()
+
+    ps = PRESETS[args.preset]
+    if args.nodes is None:
+        args.nodes = ps["nodes"]
+    if args.edges is None:
+        args.edges = ps["edges"]
+    if args.cache is None:
+        args.cache = ps["cache"]
+    feat_dim, n_classes, n_train = ps["feat_dim"], ps["classes"], ps["train"]
+
+    # refuse to start if the box cannot hold the host-side working set
+    # (graph indices + features + pinned cold tier) — failing cleanly beats
+    # driving the box out of memory
+    need = args.edges * 8 + 2.2 * args.nodes * feat_dim * 4 + (1 << 32)
+    try:
+        avail = next(int(l.split()[1]) * 1024
+                     for l in open("/proc/meminfo")
+                     if l.startswith("MemAvailable"))
+    except (OSError, StopIteration):
+        avail = None
+    if avail is not None and avail < need:
+        raise SystemExit(f"host RAM too small for preset {args.preset}: "
+                         f"need ~{need/2**30:.0f} GiB, "
+                         f"available {avail/2**30:.0f} GiB")
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -103,7 +132,12 @@ def main():
                                     edges=args.edges)
     csr_topo = quiver.CSRTopo(indptr=indptr, indices=indices)
     g = torch.Generator().manual_seed(0)
-    feat_cpu = torch.randn(args.nodes, FEAT_DIM, generator=g)
+    # chunked in-place fill: torch.randn of a 57 GB papers100M feature
+    # tensor would take minutes; uniform_ is one pass
+    feat_cpu = torch.empty(args.nodes, feat_dim)
+    step = max(1, (1 << 28) // max(feat_dim, 1))
+    for beg in range(0, args.nodes, step):
+        feat_cpu[beg:beg + step].uniform_(-1.0, 1.0, generator=g)
 
     if args.cache_policy == "p2p_clique_replicate":
         quiver.init_p2p(list(range(world)))
@@ -118,23 +152,23 @@ def main():
                              csr_topo=csr_topo)
     feature.from_cpu_tensor(feat_cpu)
 
-    model = GraphSAGE(FEAT_DIM, HIDDEN, N_CLASSES, num_layers=len(FANOUT),
+    model = GraphSAGE(feat_dim, HIDDEN, n_classes, num_layers=len(FANOUT),
                       dropout=0.0).to(device)
     if distributed:
         model = torch.nn.parallel.DistributedDataParallel(
             model, device_ids=[local_rank])
     opt = torch.optim.Adam(model.parameters(), lr=3e-3)
     # labels resident on GPU: indexing them with n_id stays device-side
-    y = torch.randint(0, N_CLASSES, (args.nodes,), generator=g).to(device)
+    y = torch.randint(0, n_classes, (args.nodes,), generator=g).to(device)
 
     # per-rank seed stream: train nodes are a fixed uniform draw over ALL
     # nodes (real train_idx is not degree-biased; hot-head-only seeds would
     # overstate cache hit rate)
     tg = torch.Generator().manual_seed(42)
-    train_idx = torch.randint(0, args.nodes, (N_TRAIN,), generator=tg)
+    train_idx = torch.randint(0, args.nodes, (n_train,), generator=tg)
     sg = torch.Generator().manual_seed(1234 + rank)
     n_batches = args.warmup + args.steps
-    batches = [train_idx[torch.randint(0, N_TRAIN, (args.batch,),
+    batches = [train_idx[torch.randint(0, n_train, (args.batch,),
                                        generator=sg)]
                for _ in range(n_batches)]
     if rank == 0:
@@ -187,14 +221,15 @@ def main():
 
     if rank == 0:
         ms_per_step = elapsed / args.steps * 1000
-        steps_per_epoch = (N_TRAIN + args.batch * world - 1) // (args.batch *
+        steps_per_epoch = (n_train + args.batch * world - 1) // (args.batch *
                                                                  world)
         epoch_seconds = ms_per_step / 1000 * steps_per_epoch
-        ref = REF_EPOCH_SECONDS.get(world)
+        ref = (REF_EPOCH_SECONDS.get(world)
+               if args.preset == "products" else None)
         vs_baseline = (ref / epoch_seconds) if ref else None
         print(json.dumps({
             "metric": "train-sampled-edges/sec (GraphSAGE e2e step, "
-                      "ogbn-products-shaped synthetic)",
+                      f"ogbn-{args.preset}-shaped synthetic)",
             "value": edges_done / elapsed,
             "unit": "edges/s",
             "n_gpus": world,
@@ -205,11 +240,12 @@ def main():
             "scaling": "weak",
             "vs_baseline": vs_baseline,
             "dtype": "fp32",
-            "data": "synthetic (ogbn-products shape: 2.45M nodes, 123.7M "
-                    "edges, 100 feats, power-law degrees), random-init "
+            "data": f"synthetic (ogbn-{args.preset} shape: "
+                    f"{args.nodes/1e6:.2f}M nodes, {m/1e6:.1f}M edges, "
+                    f"{feat_dim} feats, power-law degrees), random-init "
                     "3-layer SAGE",
             "config": {
-                "model": "graphsage-3L-h256 ogbn-products",
+                "model": f"graphsage-3L-h{HIDDEN} ogbn-{args.preset}",
                 "global_batch": args.batch * world,
                 "fanout": FANOUT,
                 "parallelism": f"dp{world}",
