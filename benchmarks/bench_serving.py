@@ -188,11 +188,33 @@ def main():
         c.join()
     batcher.stop()
 
+    def dump_worker_stacks():
+        """Hang diagnosis: SIGUSR1 -> faulthandler stack dump to stderr."""
+        import signal
+        pids = [p.pid for p in batcher.procs]
+        if sampler_pool is not None:
+            pids += [p.pid for p in sampler_pool.procs]
+        ctx = getattr(server, "spawn_ctx", None)
+        if ctx is not None:
+            pids += [p.pid for p in ctx.processes]
+        for pid in pids:
+            try:
+                os.kill(pid, signal.SIGUSR1)
+            except (OSError, TypeError):
+                pass
+        time.sleep(3)
+
     stats = []
     deadline = time.time() + 120
     for q in server.result_queue_list():
         while time.time() < deadline:
-            item = q.get(timeout=120)
+            try:
+                item = q.get(timeout=120)
+            except Exception:
+                print("# TIMEOUT waiting for worker stats — dumping worker "
+                      "stacks to stderr", flush=True)
+                dump_worker_stacks()
+                raise
             if isinstance(item, dict):
                 stats.append(item)
                 break

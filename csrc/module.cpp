@@ -490,9 +490,50 @@ class ShardTensor {
             shape, torch::TensorOptions().dtype(dtype_).device(
                        torch::Device(torch::kCUDA, dev)));
         auto spec = build_spec(dev);
-        auto stream = c10::hip::getCurrentHIPStream(dev).stream();
-        qk::launch_gather(stream, spec, indices.data_ptr<int64_t>(), n,
-                          (char*)out.data_ptr());
+        auto cur = c10::hip::getCurrentHIPStream(dev);
+
+        // Tier-split: a mixed gather would run at the capped host grid for
+        // ALL rows (the cap keeps the PCIe-latency-bound zero-copy pass
+        // from hogging CUs).  Split instead: pinned-host rows on the
+        // current stream (long pole), HBM/xGMI rows concurrently on a side
+        // stream at full grid.
+        uint32_t host_mask = 0;
+        for (int i = 0; i < spec.nshards; ++i)
+            if (shards_[i].device < 0) host_mask |= (1u << i);
+        host_mask &= spec.access_mask;
+        uint32_t dev_mask = spec.access_mask & ~host_mask;
+
+        if (host_mask != 0 && dev_mask != 0) {
+            qk::GatherSpec hs = spec;
+            hs.access_mask = host_mask;
+            hs.has_host_shard = true;
+            qk::GatherSpec ds = spec;
+            ds.access_mask = dev_mask;
+            ds.has_host_shard = false;
+            auto side = c10::hip::getStreamFromPool(false, dev);
+            auto& ev = split_events_[dev];  // events live on `dev`
+            if (!ev.first) {
+                QK_CHECK_HIP(hipEventCreateWithFlags(
+                    &ev.first, hipEventDisableTiming));
+                QK_CHECK_HIP(hipEventCreateWithFlags(
+                    &ev.second, hipEventDisableTiming));
+            }
+            QK_CHECK_HIP(hipEventRecord(ev.first, cur.stream()));
+            QK_CHECK_HIP(hipStreamWaitEvent(side.stream(), ev.first, 0));
+            qk::launch_gather(cur.stream(), hs, indices.data_ptr<int64_t>(),
+                              n, (char*)out.data_ptr());
+            qk::launch_gather(side.stream(), ds, indices.data_ptr<int64_t>(),
+                              n, (char*)out.data_ptr());
+            QK_CHECK_HIP(hipEventRecord(ev.second, side.stream()));
+            QK_CHECK_HIP(hipStreamWaitEvent(cur.stream(), ev.second, 0));
+            // caching-allocator hazard: out/indices are used on `side`
+            out.record_stream(side.unwrap());
+            indices.record_stream(side.unwrap());
+        } else {
+            qk::launch_gather(cur.stream(), spec,
+                              indices.data_ptr<int64_t>(), n,
+                              (char*)out.data_ptr());
+        }
         return out;
     }
 
@@ -602,6 +643,10 @@ class ShardTensor {
     torch::Dtype dtype_ = torch::kFloat32;
     int64_t row_bytes_ = 0;
     bool row_shape_init_ = false;
+    // per-executing-device split/join events for the tier-split gather
+    // (events must live on the device whose streams record them)
+    mutable std::unordered_map<int, std::pair<hipEvent_t, hipEvent_t>>
+        split_events_;
 };
 
 // ---------------------------------------------------------------------------

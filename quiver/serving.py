@@ -25,6 +25,17 @@ class _Stop(object):
     pass
 
 
+def _enable_stack_dump():
+    """SIGUSR1 -> dump this worker's python stacks to stderr (hang
+    diagnosis: `kill -USR1 <pid>` or bench_serving's timeout handler)."""
+    try:
+        import faulthandler
+        import signal
+        faulthandler.register(signal.SIGUSR1)
+    except (ImportError, AttributeError, ValueError):
+        pass
+
+
 def _pin(cpu_range, idx):
     if cpu_range:
         try:
@@ -64,6 +75,7 @@ class RequestBatcher(object):
             self.procs.append(p)
 
     def fixed_despatch(self, idx):
+        _enable_stack_dump()
         _pin(self.cpu_range, idx)
         stream_queue = self.stream_queue_list[idx]
         if self.sample_mode == "CPU":
@@ -78,6 +90,7 @@ class RequestBatcher(object):
             batched_queue.put(item)
 
     def auto_despatch(self, idx):
+        _enable_stack_dump()
         _pin(self.cpu_range, idx)
         stream_queue = self.stream_queue_list[idx]
         gpu_q = self.gpu_batched_queue_list[idx % self.device_num]
@@ -134,6 +147,7 @@ class HybridSampler(object):
     def cpu_sampler_worker_loop(self, rank, sample_task_queue_list,
                                 result_queue_list, device_num, sizes,
                                 csr_topo):
+        _enable_stack_dump()
         _pin(self.cpu_range, rank)
         cpu_sampler = GraphSageSampler(csr_topo, sizes, device="cpu",
                                        mode="CPU")
@@ -199,7 +213,7 @@ class InferenceServer(object):
                                   for _ in range(self.num_proc)]
 
     def start(self, join=True):
-        mp.spawn(self.run,
+        self.spawn_ctx = mp.spawn(self.run,
                  args=(self.device_list, self.cpu_sampled_queue_list,
                        self.model_path, self.x_feature,
                        self.gpu_task_queue_list, self.sample_mode,
@@ -210,6 +224,7 @@ class InferenceServer(object):
     def run(self, rank, device_list, cpu_sampled_queue_list, model_path,
             feature, gpu_sample_task_queue_list, sample_mode, csr_topo, sizes,
             num_proc, uva_gpu, output_queue_list):
+        _enable_stack_dump()
         output_queue = output_queue_list[rank]
         if sample_mode == "Auto":
             if rank < num_proc // 2:
