@@ -22,6 +22,7 @@ KERNEL_SOURCES = [
     "reindex_kernels.hip",
     "gather_kernels.hip",
     "segment_kernels.hip",
+    "wgrad_kernels.hip",
 ]
 TORCH_SOURCES = ["module.cpp"]
 OUT = os.path.join(ROOT, "torch_quiver.so")

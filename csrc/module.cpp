@@ -684,6 +684,28 @@ torch::Tensor segment_mean_gather_backward(torch::Tensor grad_out,
     return grad_x;
 }
 
+// Tall-skinny weight gradient: (A^T @ B, optional column-sums of A).
+// A [K×M] (grad_out), B [K×N] (layer input), K = frontier size.
+std::tuple<torch::Tensor, torch::Tensor> wgrad(torch::Tensor a,
+                                               torch::Tensor b,
+                                               bool want_bias) {
+    TORCH_CHECK(a.is_cuda() && b.is_cuda() && a.dim() == 2 && b.dim() == 2 &&
+                a.size(0) == b.size(0) &&
+                a.dtype() == torch::kFloat32 &&
+                b.dtype() == torch::kFloat32,
+                "wgrad: fp32 CUDA 2-D tensors with matching K expected");
+    a = a.contiguous();
+    b = b.contiguous();
+    int64_t k = a.size(0);
+    int m = (int)a.size(1), n = (int)b.size(1);
+    auto c = torch::zeros({m, n}, a.options());
+    auto bias = want_bias ? torch::zeros({m}, a.options()) : torch::Tensor();
+    qk::launch_wgrad(current_stream(), a.data_ptr<float>(),
+                     b.data_ptr<float>(), c.data_ptr<float>(),
+                     want_bias ? bias.data_ptr<float>() : nullptr, k, m, n);
+    return {c, bias};
+}
+
 void init_p2p(const std::vector<int>& devices) {
     // On an 8x MI355X node every pair is xGMI-connected: enable the full
     // clique (reference init_p2p, quiver_feature.cu:378-421; no NVLink-style
@@ -860,6 +882,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("segment_mean_gather_backward", &segment_mean_gather_backward,
           py::call_guard<py::gil_scoped_release>());
 
+    m.def("wgrad", &wgrad,
+          "tall-skinny A^T@B (+ optional A column sums) via split-K atomics");
     m.def("create_nccl_id", &create_nccl_id);
     py::class_<RcclComm>(m, "NcclComm")
         .def(py::init<int, int, py::bytes>())

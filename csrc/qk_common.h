@@ -114,6 +114,14 @@ void launch_gather(hipStream_t s, const GatherSpec& spec,
 void launch_scatter(hipStream_t s, const GatherSpec& spec,
                     const int64_t* indices, int64_t n, const char* src);
 
+// ---------- tall-skinny weight-grad GEMM (wgrad_kernels.hip) -------------
+
+// C[M×N] += A^T @ B with A [K×M], B [K×N] row-major, K huge; C (and
+// bias_grad, = column sums of A, if non-null) must be pre-zeroed.
+// Split-K with fp32 global atomics: reduction order nondeterministic.
+void launch_wgrad(hipStream_t s, const float* A, const float* B, float* C,
+                  float* bias_grad, int64_t K, int M, int N);
+
 // ---------- fused message-passing aggregation (segment_kernels.hip) ------
 
 // out[d] = mean over edges e in [dst_ptr[d], dst_ptr[d+1]) of x[src[e]]

@@ -16,7 +16,7 @@ import torch.nn.functional as F
 
 from . import _ext
 
-__all__ = ["SAGEConv", "GATConv", "GraphSAGE", "GAT"]
+__all__ = ["SAGEConv", "GATConv", "GraphSAGE", "GAT", "QLinear"]
 
 
 def _as_pair(x):
@@ -47,6 +47,48 @@ class _SegmentMeanAgg(torch.autograd.Function):
         return grad_x, None, None
 
 
+class _QLinearFn(torch.autograd.Function):
+    """Linear layer whose weight gradient uses the split-K HIP kernel
+    (csrc/wgrad_kernels.hip).
+
+    The wgrad GEMM here is tall-skinny — C[out×in] = grad_out^T @ x with
+    K = frontier rows (10^4..10^6) — a shape rocBLAS stream-K runs ~60x
+    off the HBM roof on gfx950.  Forward and data-grad GEMMs keep rocBLAS
+    (their shapes are fine).  Bias grad is folded into the same kernel
+    pass.  fp32 atomic split-K: wgrad is bitwise-nondeterministic (same
+    class of nondeterminism as rocBLAS GSU).
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ctx.save_for_backward(x, weight)
+        ctx.has_bias = bias is not None
+        return F.linear(x, weight, bias)
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        x, weight = ctx.saved_tensors
+        grad_out = grad_out.contiguous()
+        grad_x = grad_out @ weight if ctx.needs_input_grad[0] else None
+        grad_w, grad_b = _ext.wgrad(grad_out, x, ctx.has_bias)
+        return grad_x, grad_w, (grad_b if ctx.has_bias else None)
+
+
+# frontier sizes below this use plain F.linear (rocBLAS wgrad is fine there)
+_WGRAD_MIN_K = 16384
+
+
+class QLinear(nn.Linear):
+    """nn.Linear with the custom tall-skinny weight-grad path on GPU fp32
+    inputs with a large leading (frontier) dimension."""
+
+    def forward(self, x):
+        if (x.is_cuda and x.dtype == torch.float32 and x.dim() == 2
+                and x.size(0) >= _WGRAD_MIN_K and torch.is_grad_enabled()):
+            return _QLinearFn.apply(x, self.weight, self.bias)
+        return F.linear(x, self.weight, self.bias)
+
+
 def _mean_aggregate(x_src, src, dst, n_dst, sorted_dst=False):
     """Mean of x_src[src] grouped by dst.  Uses the fused HIP kernel when
     the caller guarantees dst is sorted ascending (our sampler's layout)
@@ -75,8 +117,8 @@ class SAGEConv(nn.Module):
                  sorted_dst=False):
         super().__init__()
         self.sorted_dst = sorted_dst
-        self.lin_l = nn.Linear(in_channels, out_channels, bias=bias)  # neigh
-        self.lin_r = nn.Linear(in_channels, out_channels, bias=False)  # self
+        self.lin_l = QLinear(in_channels, out_channels, bias=bias)  # neigh
+        self.lin_r = QLinear(in_channels, out_channels, bias=False)  # self
         self.reset_parameters()
 
     def reset_parameters(self):
@@ -92,8 +134,14 @@ class SAGEConv(nn.Module):
         # aggregate-then-project: segment mean on in_channels, one GEMM after
         agg = _mean_aggregate(x_src, src, dst, n_dst,
                               sorted_dst=self.sorted_dst)
-        # self path fused into the neighbor GEMM's epilogue via addmm
         out = self.lin_l(agg)
+        if (x_dst.is_cuda and x_dst.dtype == torch.float32
+                and x_dst.size(0) >= _WGRAD_MIN_K
+                and torch.is_grad_enabled()):
+            # large frontier: route the self path through QLinear so ITS
+            # weight grad also takes the split-K kernel
+            return out + self.lin_r(x_dst)
+        # small/no-grad: self path fused into the neighbor GEMM via addmm
         return out.addmm_(x_dst, self.lin_r.weight.t()) if not out.requires_grad \
             else torch.addmm(out, x_dst, self.lin_r.weight.t())
 

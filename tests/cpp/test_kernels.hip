@@ -474,6 +474,44 @@ void test_segment_mean() {
     printf("ok test_segment_mean\n");
 }
 
+void test_wgrad() {
+    std::mt19937 rng(7);
+    const int64_t K = 30000;
+    const int M = 100, N = 130;  // non-multiples of the 64x64 tile
+    std::vector<float> a(K * M), b(K * N);
+    std::uniform_real_distribution<float> ud(-1.f, 1.f);
+    for (auto& v : a) v = ud(rng);
+    for (auto& v : b) v = ud(rng);
+    auto* d_a = dalloc<float>(a.size());
+    auto* d_b = dalloc<float>(b.size());
+    auto* d_c = dalloc<float>((size_t)M * N);
+    auto* d_bias = dalloc<float>(M);
+    h2d(d_a, a);
+    h2d(d_b, b);
+    QK_CHECK_HIP(hipMemset(d_c, 0, sizeof(float) * M * N));
+    QK_CHECK_HIP(hipMemset(d_bias, 0, sizeof(float) * M));
+    qk::launch_wgrad(nullptr, d_a, d_b, d_c, d_bias, K, M, N);
+    QK_CHECK_HIP(hipDeviceSynchronize());
+    auto c = d2h(d_c, (size_t)M * N);
+    auto bias = d2h(d_bias, M);
+    // CPU reference in double
+    double tol = 1e-4 * std::sqrt((double)K) * 10;
+    for (int mm = 0; mm < M; ++mm) {
+        double bs = 0;
+        for (int64_t k = 0; k < K; ++k) bs += a[k * M + mm];
+        REQUIRE(std::abs(bias[mm] - bs) < tol);
+    }
+    for (int mm = 0; mm < M; mm += 7)
+        for (int nn = 0; nn < N; ++nn) {
+            double acc = 0;
+            for (int64_t k = 0; k < K; ++k)
+                acc += (double)a[k * M + mm] * b[k * N + nn];
+            REQUIRE(std::abs(c[mm * N + nn] - acc) < tol);
+        }
+    hipFree(d_a); hipFree(d_b); hipFree(d_c); hipFree(d_bias);
+    printf("ok test_wgrad\n");
+}
+
 }  // namespace
 
 int main() {
@@ -489,6 +527,7 @@ int main() {
     test_cal_next();
     test_gather_scatter();
     test_segment_mean();
+    test_wgrad();
     printf("ALL C++ KERNEL TESTS PASSED\n");
     return 0;
 }
