@@ -9,6 +9,7 @@
 #include <torch/extension.h>
 
 #include <ATen/Parallel.h>
+#include <c10/hip/HIPCachingAllocator.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
@@ -526,9 +527,13 @@ class ShardTensor {
                               n, (char*)out.data_ptr());
             QK_CHECK_HIP(hipEventRecord(ev.second, side.stream()));
             QK_CHECK_HIP(hipStreamWaitEvent(cur.stream(), ev.second, 0));
-            // caching-allocator hazard: out/indices are used on `side`
-            out.record_stream(side.unwrap());
-            indices.record_stream(side.unwrap());
+            // caching-allocator hazard: out/indices are used on `side`.
+            // (Tensor::record_stream wants a masquerading-as-CUDA stream;
+            // go to the HIP allocator directly.)
+            c10::hip::HIPCachingAllocator::recordStream(
+                out.storage().data_ptr(), side);
+            c10::hip::HIPCachingAllocator::recordStream(
+                indices.storage().data_ptr(), side);
         } else {
             qk::launch_gather(cur.stream(), spec,
                               indices.data_ptr<int64_t>(), n,

@@ -26,13 +26,25 @@ class _Stop(object):
 
 
 def _enable_stack_dump():
-    """SIGUSR1 -> dump this worker's python stacks to stderr (hang
-    diagnosis: `kill -USR1 <pid>` or bench_serving's timeout handler)."""
+    """Worker-process init.
+
+    1. SIGUSR1 -> dump this worker's python stacks to stderr (hang
+       diagnosis: `kill -USR1 <pid>` or bench_serving's timeout handler).
+    2. CPU-tensor sharing strategy -> file_system: queue payloads here go
+       through mp.Manager proxies, and the default fd-passing strategy
+       accumulates one fd in the manager process per shared tensor — at
+       serving request rates that exhausts the fd limit and deadlocks the
+       manager (observed as workers stuck in storage._share_fd_cpu_).
+    """
     try:
         import faulthandler
         import signal
         faulthandler.register(signal.SIGUSR1)
     except (ImportError, AttributeError, ValueError):
+        pass
+    try:
+        mp.set_sharing_strategy("file_system")
+    except (RuntimeError, ValueError):
         pass
 
 
