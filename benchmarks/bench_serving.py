@@ -146,7 +146,7 @@ def main():
         else:
             threshold = 0
 
-    stream_queues = [mp.Manager().Queue() for _ in range(device_num)]
+    stream_queues = [mp.get_context("spawn").Queue() for _ in range(device_num)]
     batcher = quiver.RequestBatcher(device_num, stream_queues,
                                     input_proc_per_device=1,
                                     sample_mode=args.mode,
@@ -245,3 +245,9 @@ if __name__ == "__main__":
     # batcher/client/CPU-sampler procs fork (no CUDA touched in them);
     # InferenceServer always uses the spawn context internally.
     main()
+    # skip interpreter-exit finalizers: daemon workers may hold undrained
+    # queue items (results enqueued after a _Stop raced past them), and
+    # their feeder threads would block a clean exit forever
+    sys.stdout.flush()
+    sys.stderr.flush()
+    os._exit(0)

@@ -47,7 +47,7 @@ def main(n=100_000, dim=64, requests=500, batch=64, device_list=None,
     torch.save(GraphSAGE(dim, 128, 16, num_layers=2), model_path)
 
     nproc = len(device_list)
-    stream_queues = [mp.Manager().Queue() for _ in range(nproc)]
+    stream_queues = [mp.get_context("spawn").Queue() for _ in range(nproc)]
     batcher = quiver.RequestBatcher(device_num=nproc,
                                     stream_queue_list=stream_queues,
                                     input_proc_per_device=1,

@@ -20,6 +20,12 @@ from .pyg import GraphSageSampler
 __all__ = ["RequestBatcher", "HybridSampler", "InferenceServer",
            "InferenceServer_Debug"]
 
+# All pipeline queues come from the spawn context: a fork-context mp.Queue
+# passed into an mp.spawn child NEVER delivers (the child blocks in get()
+# silently).  Spawn-context queues work for both forked (batcher/sampler)
+# and spawned (inference) processes.
+_ctx = mp.get_context("spawn")
+
 
 class _Stop(object):
     pass
@@ -65,9 +71,14 @@ class RequestBatcher(object):
                  sample_mode="GPU", request_mode="CPU", threshold=800,
                  batch_time_limit=10, fixed_batch_size=512,
                  neighbour_path=None, cpu_range=[]):
-        self.cpu_batched_queue_list = [mp.Manager().Queue()
+        # direct (spawn-context) queues, NOT mp.Manager().Queue(): tensor
+        # payloads through a manager proxy funnel every put/get through the
+        # manager server process, which wedges under serving request rates.
+        # Direct queues pass the shm handle endpoint-to-endpoint like torch
+        # DataLoader workers.
+        self.cpu_batched_queue_list = [_ctx.Queue()
                                        for _ in range(device_num)]
-        self.gpu_batched_queue_list = [mp.Manager().Queue()
+        self.gpu_batched_queue_list = [_ctx.Queue()
                                        for _ in range(device_num)]
         self.stream_queue_list = stream_queue_list
         self.sample_mode = sample_mode
@@ -142,7 +153,7 @@ class HybridSampler(object):
         self.sizes = sizes
         self.cpu_batched_queue_list = batched_queue_list[0]
         self.gpu_batched_queue_list = batched_queue_list[1]
-        self.cpu_sampled_queue_list = [mp.Manager().Queue()
+        self.cpu_sampled_queue_list = [_ctx.Queue()
                                        for _ in range(device_num)]
         self.procs = []
 
@@ -221,7 +232,7 @@ class InferenceServer(object):
         self.proc_num_per_device = proc_num_per_device
         self.uva_gpu = uva_gpu
         self.num_proc = len(self.device_list) * self.proc_num_per_device
-        self.output_queue_list = [mp.Manager().Queue()
+        self.output_queue_list = [_ctx.Queue()
                                   for _ in range(self.num_proc)]
 
     def start(self, join=True):
@@ -237,6 +248,20 @@ class InferenceServer(object):
             feature, gpu_sample_task_queue_list, sample_mode, csr_topo, sizes,
             num_proc, uva_gpu, output_queue_list):
         _enable_stack_dump()
+        try:
+            self._run(rank, device_list, cpu_sampled_queue_list, model_path,
+                      feature, gpu_sample_task_queue_list, sample_mode,
+                      csr_topo, sizes, num_proc, uva_gpu, output_queue_list)
+        except Exception:
+            # started with join=False the spawn context is never joined, so
+            # surface worker failures on stderr instead of dying silently
+            import traceback
+            traceback.print_exc()
+            raise
+
+    def _run(self, rank, device_list, cpu_sampled_queue_list, model_path,
+             feature, gpu_sample_task_queue_list, sample_mode, csr_topo,
+             sizes, num_proc, uva_gpu, output_queue_list):
         output_queue = output_queue_list[rank]
         if sample_mode == "Auto":
             if rank < num_proc // 2:

@@ -54,7 +54,7 @@ def test_serving_gpu_pipeline(tmp_path):
     model_path = str(tmp_path / "m.pt")
     torch.save(GraphSAGE(16, 32, 4, num_layers=2, dropout=0.0), model_path)
 
-    stream_queues = [mp.Manager().Queue()]
+    stream_queues = [mp.get_context("spawn").Queue()]
     batcher = quiver.RequestBatcher(device_num=1,
                                     stream_queue_list=stream_queues,
                                     input_proc_per_device=1,

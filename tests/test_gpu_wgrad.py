@@ -18,13 +18,14 @@ def test_wgrad_matches_torch(k, m, n):
     a = torch.randn(k, m, device="cuda", generator=g)
     b = torch.randn(k, n, device="cuda", generator=g)
     c, bias = _ext.wgrad(a, b, True)
-    want_c = a.t() @ b
-    want_bias = a.sum(0)
-    # fp32 atomic split-K: tolerance scales with sqrt(K)
-    tol = 1e-5 * max(k, 1) ** 0.5
-    assert torch.allclose(c, want_c, atol=tol, rtol=1e-4), \
-        (c - want_c).abs().max().item()
-    assert torch.allclose(bias, want_bias, atol=tol, rtol=1e-4)
+    # fp64 ground truth; fp32 split-K accumulation error grows ~sqrt(K),
+    # so judge by relative error against the fp64 result
+    want_c = (a.double().t() @ b.double())
+    want_bias = a.double().sum(0)
+    rel_c = (c.double() - want_c).norm() / want_c.norm()
+    rel_b = (bias.double() - want_bias).norm() / max(want_bias.norm(), 1e-30)
+    assert rel_c < 1e-5, float(rel_c)
+    assert rel_b < 1e-4, float(rel_b)  # bias sums can cancel toward 0
 
 
 def test_wgrad_no_bias():

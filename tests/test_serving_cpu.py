@@ -28,7 +28,7 @@ def test_cpu_serving_pipeline(serving_graph, tmp_path):
     model_path = str(tmp_path / "model.pt")
     torch.save(model, model_path)
 
-    stream_queues = [mp.Manager().Queue()]
+    stream_queues = [mp.get_context("spawn").Queue()]
     batcher = quiver.RequestBatcher(device_num=1,
                                     stream_queue_list=stream_queues,
                                     input_proc_per_device=1,
@@ -65,7 +65,7 @@ def test_auto_routing_threshold(tmp_path):
     path = str(tmp_path / "nbr.npy")
     np.save(path, neighbour)
 
-    stream_queues = [mp.Manager().Queue()]
+    stream_queues = [mp.get_context("spawn").Queue()]
     batcher = quiver.RequestBatcher(device_num=1,
                                     stream_queue_list=stream_queues,
                                     input_proc_per_device=1,
