@@ -33,9 +33,17 @@ constexpr int kWaveSize = 64;  // CDNA4 wavefront
 // ---------- sampling (sample_kernels.hip) ----------
 
 // counts[i] = degree(seeds[i]); capped[i] = min(counts[i], k). k<0 => no cap.
+//
+// Dynamic-count convention (here and below): `n`/`n_nbrs` are HOST upper
+// bounds used for grid sizing and slack zeroing; when `n_dev`/`m_dev` is
+// non-null the exact count lives on the DEVICE (written by a prior kernel
+// in the same stream) and the kernel reads it — this is what lets the
+// fused multi-hop sampling loop run an entire batch without a single
+// host-device synchronization (slack elements of count/flag buffers are
+// zeroed so downstream scans stay exact).
 void launch_capped_degree(hipStream_t s, const int64_t* indptr,
                           const int64_t* seeds, int64_t n, int k,
-                          int64_t* capped);
+                          int64_t* capped, const int64_t* n_dev = nullptr);
 
 // Exclusive scan over n int64 values; also writes total = sum to d_total.
 // temp_bytes(n) tells the caller how much scratch to allocate.
@@ -51,7 +59,8 @@ void launch_exclusive_scan(hipStream_t s, void* temp, size_t temp_bytes,
 void launch_sample(hipStream_t s, const int64_t* indptr, const int64_t* indices,
                    const int64_t* eid_base, const int64_t* seeds, int64_t n,
                    int k, const int64_t* prefix, int64_t* out_nbrs,
-                   int64_t* out_eids, uint64_t rng_seed);
+                   int64_t* out_eids, uint64_t rng_seed,
+                   const int64_t* n_dev = nullptr);
 
 // Access-probability propagation (one hop):
 // cur[v] = 1 - (1 - last[v]) * prod_{u in N(v)} (1 - last[u]*min(1, k/deg(u)))
@@ -71,23 +80,30 @@ void launch_reindex_init(hipStream_t s, int64_t* keys, int32_t* pos,
                          int64_t capacity);
 void launch_hash_insert(hipStream_t s, int64_t* keys, int32_t* pos,
                         int64_t capacity, const int64_t* seeds, int64_t n_seeds,
-                        const int64_t* nbrs, int64_t n_nbrs);
+                        const int64_t* nbrs, int64_t n_nbrs,
+                        const int64_t* n_seeds_dev = nullptr,
+                        const int64_t* n_nbrs_dev = nullptr);
 void launch_mark_first(hipStream_t s, const int64_t* keys, const int32_t* pos,
                        int64_t capacity, const int64_t* seeds, int64_t n_seeds,
-                       const int64_t* nbrs, int64_t n_nbrs, int64_t* flags);
+                       const int64_t* nbrs, int64_t n_nbrs, int64_t* flags,
+                       const int64_t* n_seeds_dev = nullptr,
+                       const int64_t* n_nbrs_dev = nullptr);
 void launch_compact_unique(hipStream_t s, const int64_t* keys, int32_t* local,
                            const int32_t* pos, int64_t capacity,
                            const int64_t* seeds, int64_t n_seeds,
                            const int64_t* nbrs, int64_t n_nbrs,
                            const int64_t* scanned_flags, const int64_t* flags,
-                           int64_t* out_nodes);
+                           int64_t* out_nodes,
+                           const int64_t* n_seeds_dev = nullptr,
+                           const int64_t* n_nbrs_dev = nullptr);
 void launch_lookup_local(hipStream_t s, const int64_t* keys,
                          const int32_t* local, int64_t capacity,
-                         const int64_t* nbrs, int64_t n_nbrs, int64_t* col_idx);
+                         const int64_t* nbrs, int64_t n_nbrs, int64_t* col_idx,
+                         const int64_t* n_nbrs_dev = nullptr);
 // row_idx[prefix[i]+j] = i  for j < counts[i]
 void launch_expand_rows(hipStream_t s, const int64_t* prefix,
                         const int64_t* counts, int64_t n_seeds,
-                        int64_t* row_idx);
+                        int64_t* row_idx, const int64_t* n_dev = nullptr);
 
 // ---------- feature gather (gather_kernels.hip) ----------
 

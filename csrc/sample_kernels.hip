@@ -38,10 +38,16 @@ __device__ __forceinline__ int64_t bounded_rand(uint64_t h, int64_t bound) {
 __global__ void capped_degree_kernel(const int64_t* __restrict__ indptr,
                                      const int64_t* __restrict__ seeds,
                                      int64_t n, int k,
-                                     int64_t* __restrict__ capped) {
+                                     int64_t* __restrict__ capped,
+                                     const int64_t* __restrict__ n_dev) {
+    const int64_t nn = n_dev ? *n_dev : n;
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < n; i += stride) {
+        if (i >= nn) {          // slack: keep the downstream scan exact
+            capped[i] = 0;
+            continue;
+        }
         int64_t v = seeds[i];
         int64_t deg = indptr[v + 1] - indptr[v];
         capped[i] = (k >= 0 && deg > k) ? k : deg;
@@ -56,16 +62,17 @@ sample_kernel(const int64_t* __restrict__ indptr,
               const int64_t* __restrict__ seeds, int64_t n, int k,
               const int64_t* __restrict__ prefix,
               int64_t* __restrict__ out_nbrs, int64_t* __restrict__ out_eids,
-              uint64_t rng_seed) {
+              uint64_t rng_seed, const int64_t* __restrict__ n_dev) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     int* slots = reinterpret_cast<int*>(smem);  // ROWS_PER_BLOCK * k
 
+    const int64_t nn = n_dev ? *n_dev : n;
     const int sub_id = threadIdx.x / SUB;
     const int lane = threadIdx.x % SUB;
     int64_t row = (int64_t)blockIdx.x * ROWS_PER_BLOCK + sub_id;
     const int64_t row_stride = (int64_t)gridDim.x * ROWS_PER_BLOCK;
 
-    for (; row < n; row += row_stride) {
+    for (; row < nn; row += row_stride) {
         const int64_t v = seeds[row];
         const int64_t beg = indptr[v];
         const int64_t deg = indptr[v + 1] - beg;
@@ -166,10 +173,10 @@ inline int grid_for(int64_t work_items, int per_block) {
 
 void launch_capped_degree(hipStream_t s, const int64_t* indptr,
                           const int64_t* seeds, int64_t n, int k,
-                          int64_t* capped) {
+                          int64_t* capped, const int64_t* n_dev) {
     if (n == 0) return;
-    capped_degree_kernel<<<grid_for(n, BLOCK), BLOCK, 0, s>>>(indptr, seeds, n,
-                                                              k, capped);
+    capped_degree_kernel<<<grid_for(n, BLOCK), BLOCK, 0, s>>>(
+        indptr, seeds, n, k, capped, n_dev);
     QK_CHECK_HIP(hipGetLastError());
 }
 
@@ -201,7 +208,8 @@ void launch_exclusive_scan(hipStream_t s, void* temp, size_t temp_bytes,
 void launch_sample(hipStream_t s, const int64_t* indptr, const int64_t* indices,
                    const int64_t* eid_base, const int64_t* seeds, int64_t n,
                    int k, const int64_t* prefix, int64_t* out_nbrs,
-                   int64_t* out_eids, uint64_t rng_seed) {
+                   int64_t* out_eids, uint64_t rng_seed,
+                   const int64_t* n_dev) {
     if (n == 0) return;
     if (k < 1) throw std::runtime_error("sample: fanout k must be >= 1");
     size_t lds = (size_t)ROWS_PER_BLOCK * k * sizeof(int);
@@ -212,12 +220,12 @@ void launch_sample(hipStream_t s, const int64_t* indptr, const int64_t* indices,
         sample_kernel<true><<<grid, BLOCK, lds, s>>>(indptr, indices, eid_base,
                                                      seeds, n, k, prefix,
                                                      out_nbrs, out_eids,
-                                                     rng_seed);
+                                                     rng_seed, n_dev);
     else
         sample_kernel<false><<<grid, BLOCK, lds, s>>>(indptr, indices, eid_base,
                                                       seeds, n, k, prefix,
                                                       out_nbrs, nullptr,
-                                                      rng_seed);
+                                                      rng_seed, n_dev);
     QK_CHECK_HIP(hipGetLastError());
 }
 
