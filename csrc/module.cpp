@@ -624,7 +624,22 @@ class ShardTensor {
         host_mask &= spec.access_mask;
         uint32_t dev_mask = spec.access_mask & ~host_mask;
 
-        if (host_mask != 0 && dev_mask != 0) {
+        if (host_mask != 0 && dev_mask != 0 && staged_wanted(n)) {
+            // CPU-staged host tier: zero-copy kernel reads of scattered
+            // 400 B rows top out near ~11 GB/s (64 B uncached PCIe reads,
+            // latency-bound).  Instead: HBM/xGMI rows via the full-grid
+            // kernel (async), indices D2H, CPU threads gather host rows
+            // into pinned staging, ONE large H2D burst, then an
+            // HBM-to-HBM scatter places them.  Large DMA bursts run at
+            // full PCIe write bandwidth.
+            qk::GatherSpec ds = spec;
+            ds.access_mask = dev_mask;
+            ds.has_host_shard = false;
+            qk::launch_gather(cur.stream(), ds, indices.data_ptr<int64_t>(),
+                              n, (char*)out.data_ptr());
+            gather_host_staged(dev, cur.stream(), indices, n, spec,
+                               host_mask, out);
+        } else if (host_mask != 0 && dev_mask != 0) {
             qk::GatherSpec hs = spec;
             hs.access_mask = host_mask;
             hs.has_host_shard = true;
@@ -670,6 +685,110 @@ class ShardTensor {
         qk::launch_scatter(current_stream(), spec,
                            indices.data_ptr<int64_t>(), indices.numel(),
                            (const char*)src.data_ptr());
+    }
+
+    static bool staged_wanted(int64_t n) {
+        static int mode = [] {
+            const char* e = getenv("QUIVER_STAGED_GATHER");
+            return e ? atoi(e) : -1;  // -1 = auto
+        }();
+        if (mode == 0) return false;
+        if (mode == 1) return true;
+        return n >= 16384;  // auto: large training gathers only
+    }
+
+    // CPU-staged gather of the pinned-host tier (see gather_on).
+    void gather_host_staged(int dev, hipStream_t stream, torch::Tensor indices,
+                            int64_t n, const qk::GatherSpec& spec,
+                            uint32_t host_mask, torch::Tensor& out) {
+        const int64_t rb = row_bytes_;
+        // host base pointer + global row range of every host shard
+        struct HostShard { int64_t beg, end; const char* base; };
+        std::vector<HostShard> hosts;
+        for (int s = 0; s < spec.nshards; ++s) {
+            if (!((host_mask >> s) & 1u)) continue;
+            int64_t beg = s == 0 ? 0 : spec.ends[s - 1];
+            hosts.push_back({beg, spec.ends[s],
+                             (const char*)shards_[s].keeper.data_ptr()});
+        }
+        if (hosts.empty()) return;
+
+        auto pin_opts = torch::TensorOptions()
+                            .dtype(torch::kUInt8)
+                            .device(torch::kCPU)
+                            .pinned_memory(true);
+        if (!idx_pin_.defined() || idx_pin_.numel() < n * 8)
+            idx_pin_ = torch::empty({n * 8}, pin_opts);
+        QK_CHECK_HIP(hipMemcpyAsync(idx_pin_.data_ptr(),
+                                    indices.data_ptr<int64_t>(), n * 8,
+                                    hipMemcpyDeviceToHost, stream));
+        // waits for the D2H AND the already-launched device-tier kernel
+        QK_CHECK_HIP(hipStreamSynchronize(stream));
+        const int64_t* idx = (const int64_t*)idx_pin_.data_ptr();
+
+        // parallel count -> chunk prefix -> parallel fill+copy
+        const int64_t grain = 4096;
+        const int64_t nchunk = (n + grain - 1) / grain;
+        std::vector<int64_t> chunk_cnt(nchunk, 0);
+        auto host_of = [&hosts](int64_t v) -> const HostShard* {
+            for (const auto& h : hosts)
+                if (v >= h.beg && v < h.end) return &h;
+            return nullptr;
+        };
+        at::parallel_for(0, nchunk, 1, [&](int64_t c0, int64_t c1) {
+            for (int64_t c = c0; c < c1; ++c) {
+                int64_t cnt = 0;
+                int64_t e = std::min(n, (c + 1) * grain);
+                for (int64_t i = c * grain; i < e; ++i)
+                    if (host_of(idx[i])) ++cnt;
+                chunk_cnt[c] = cnt;
+            }
+        });
+        std::vector<int64_t> chunk_off(nchunk + 1, 0);
+        for (int64_t c = 0; c < nchunk; ++c)
+            chunk_off[c + 1] = chunk_off[c] + chunk_cnt[c];
+        const int64_t m = chunk_off[nchunk];
+        if (m == 0) return;
+        if (!pos_pin_.defined() || pos_pin_.numel() < m * 8)
+            pos_pin_ = torch::empty({std::max(m, n) * 8}, pin_opts);
+        if (!stage_pin_.defined() || stage_pin_.numel() < m * rb)
+            stage_pin_ = torch::empty({std::max(m * rb, n * rb)}, pin_opts);
+        int64_t* pos = (int64_t*)pos_pin_.data_ptr();
+        char* stage = (char*)stage_pin_.data_ptr();
+        at::parallel_for(0, nchunk, 1, [&](int64_t c0, int64_t c1) {
+            for (int64_t c = c0; c < c1; ++c) {
+                int64_t j = chunk_off[c];
+                int64_t e = std::min(n, (c + 1) * grain);
+                for (int64_t i = c * grain; i < e; ++i) {
+                    const HostShard* h = host_of(idx[i]);
+                    if (!h) continue;
+                    memcpy(stage + j * rb, h->base + (idx[i] - h->beg) * rb,
+                           rb);
+                    pos[j++] = i;
+                }
+            }
+        });
+
+        // one big burst up + positions, then HBM->HBM scatter into `out`
+        auto dev_opts = torch::TensorOptions()
+                            .dtype(torch::kUInt8)
+                            .device(torch::Device(torch::kCUDA, dev));
+        auto stage_dev = torch::empty({m * rb}, dev_opts);
+        auto pos_dev = torch::empty({m * 8}, dev_opts);
+        QK_CHECK_HIP(hipMemcpyAsync(stage_dev.data_ptr(), stage, m * rb,
+                                    hipMemcpyHostToDevice, stream));
+        QK_CHECK_HIP(hipMemcpyAsync(pos_dev.data_ptr(), pos, m * 8,
+                                    hipMemcpyHostToDevice, stream));
+        qk::GatherSpec outspec{};
+        outspec.ptrs[0] = (const char*)out.data_ptr();
+        outspec.ends[0] = n;
+        outspec.access_mask = 1u;
+        outspec.nshards = 1;
+        outspec.row_bytes = rb;
+        outspec.has_host_shard = false;
+        qk::launch_scatter(stream, outspec,
+                           (const int64_t*)pos_dev.data_ptr(), m,
+                           (const char*)stage_dev.data_ptr());
     }
 
     // Which shards are directly readable from `dev` (bit per shard).
@@ -772,6 +891,8 @@ class ShardTensor {
     // (events must live on the device whose streams record them)
     mutable std::unordered_map<int, std::pair<hipEvent_t, hipEvent_t>>
         split_events_;
+    // pinned buffers for the CPU-staged host-tier gather
+    torch::Tensor idx_pin_, pos_pin_, stage_pin_;
 };
 
 // ---------------------------------------------------------------------------
