@@ -85,7 +85,7 @@ def main():
     p.add_argument("--devices", type=int, nargs="*", default=None,
                    help="GPU ids for inference workers (default: all)")
     p.add_argument("--procs-per-device", type=int, default=1)
-    p.add_argument("--cpu-workers-per-device", type=int, default=2)
+    p.add_argument("--cpu-workers-per-device", type=int, default=8)
     p.add_argument("--nodes", type=int, default=N_NODES)
     p.add_argument("--edges", type=int, default=N_EDGES)
     p.add_argument("--threshold", type=int, default=None,

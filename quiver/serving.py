@@ -76,9 +76,12 @@ class RequestBatcher(object):
         # manager server process, which wedges under serving request rates.
         # Direct queues pass the shm handle endpoint-to-endpoint like torch
         # DataLoader workers.
-        self.cpu_batched_queue_list = [_ctx.Queue()
+        # bounded: overload applies backpressure to the stream queues
+        # instead of building an unbounded in-flight backlog that makes
+        # shutdown drain time unbounded
+        self.cpu_batched_queue_list = [_ctx.Queue(maxsize=256)
                                        for _ in range(device_num)]
-        self.gpu_batched_queue_list = [_ctx.Queue()
+        self.gpu_batched_queue_list = [_ctx.Queue(maxsize=256)
                                        for _ in range(device_num)]
         self.stream_queue_list = stream_queue_list
         self.sample_mode = sample_mode
@@ -153,7 +156,7 @@ class HybridSampler(object):
         self.sizes = sizes
         self.cpu_batched_queue_list = batched_queue_list[0]
         self.gpu_batched_queue_list = batched_queue_list[1]
-        self.cpu_sampled_queue_list = [_ctx.Queue()
+        self.cpu_sampled_queue_list = [_ctx.Queue(maxsize=256)
                                        for _ in range(device_num)]
         self.procs = []
 
@@ -188,6 +191,27 @@ class HybridSampler(object):
 
     def sampled_request_queue_list(self):
         return [self.cpu_sampled_queue_list, self.gpu_batched_queue_list]
+
+
+def _drain_after_stop(q, stop_item):
+    """The cpu_sampled queue has multiple producers; a _Stop from one
+    sampler worker can overtake results another worker already queued.
+    After seeing _Stop, keep consuming briefly so queued results are not
+    dropped; keep the sentinel circulating for sibling consumers."""
+    import queue as _queue
+    stops = 0
+    while stops < 5:
+        try:
+            item = q.get(timeout=0.5)
+        except _queue.Empty:
+            break
+        if isinstance(item, _Stop):
+            q.put(item)
+            stops += 1
+            time.sleep(0.05)
+            continue
+        yield item
+    q.put(stop_item)
 
 
 def _feature_rows(feature, n_id, device):
@@ -314,18 +338,22 @@ class InferenceServer(object):
         q = cpu_sampled_queue_list[rank % len(device_list)]
         model = _load_model(model_path, device)
         model.eval()
+        def infer(item):
+            n_id, batch_size, adjs = item[0]
+            adjs = [adj.to(device) for adj in adjs]
+            x_input = _feature_rows(feature, n_id, device)
+            out = model(x_input, adjs)
+            output_queue.put(out.cpu())
+
         with torch.no_grad():
             while True:
                 item = q.get()
                 if isinstance(item, _Stop):
-                    q.put(item)
+                    for extra in _drain_after_stop(q, item):
+                        infer(extra)
                     output_queue.put(item)
                     break
-                n_id, batch_size, adjs = item[0]
-                adjs = [adj.to(device) for adj in adjs]
-                x_input = _feature_rows(feature, n_id, device)
-                out = model(x_input, adjs)
-                output_queue.put(out.cpu())
+                infer(item)
 
     def result_queue_list(self):
         return self.output_queue_list
@@ -409,22 +437,27 @@ class InferenceServer_Debug(InferenceServer):
         model = _load_model(model_path, device)
         model.eval()
         result = []
+
+        def infer(item):
+            start_time = time.perf_counter()
+            n_id, batch_size, adjs = item[0]
+            adjs = [adj.to(device) for adj in adjs]
+            x_input = _feature_rows(feature, n_id, device)
+            out = model(x_input, adjs)
+            if device != "cpu":
+                torch.cuda.synchronize()
+            end_time = time.perf_counter()
+            result.append([start_time, start_time, end_time, batch_size,
+                           n_id.shape[0]])
+
         with torch.no_grad():
             while True:
                 item = q.get()
                 if isinstance(item, _Stop):
-                    q.put(item)
+                    for extra in _drain_after_stop(q, item):
+                        infer(extra)
                     stats = self._report(result, self.ignord_length, "CPU",
                                          rank, self.result_path, self.exp_id)
                     output_queue.put(stats if stats is not None else item)
                     break
-                start_time = time.perf_counter()
-                n_id, batch_size, adjs = item[0]
-                adjs = [adj.to(device) for adj in adjs]
-                x_input = _feature_rows(feature, n_id, device)
-                out = model(x_input, adjs)
-                if device != "cpu":
-                    torch.cuda.synchronize()
-                end_time = time.perf_counter()
-                result.append([start_time, start_time, end_time, batch_size,
-                               n_id.shape[0]])
+                infer(item)
