@@ -89,14 +89,26 @@ class QLinear(nn.Linear):
         return F.linear(x, self.weight, self.bias)
 
 
+_ARANGE_CACHE = {}
+
+
+def _arange(n, device):
+    """Cached 0..n view (avoids one alloc+launch per layer per step)."""
+    key = device.index if device.type == "cuda" else -1
+    cached = _ARANGE_CACHE.get(key)
+    if cached is None or cached.numel() < n:
+        cached = torch.arange(max(n, 1 << 20), device=device)
+        _ARANGE_CACHE[key] = cached
+    return cached[:n]
+
+
 def _mean_aggregate(x_src, src, dst, n_dst, sorted_dst=False):
     """Mean of x_src[src] grouped by dst.  Uses the fused HIP kernel when
     the caller guarantees dst is sorted ascending (our sampler's layout)
     on fp32 GPU tensors."""
     if (sorted_dst and x_src.is_cuda and x_src.dtype == torch.float32
             and dst.numel() > 0):
-        dst_ptr = torch.searchsorted(
-            dst, torch.arange(n_dst + 1, device=dst.device))
+        dst_ptr = torch.searchsorted(dst, _arange(n_dst + 1, dst.device))
         return _SegmentMeanAgg.apply(x_src, src, dst_ptr)
     agg = torch.zeros((n_dst, x_src.size(1)), dtype=x_src.dtype,
                       device=x_src.device)
@@ -233,8 +245,9 @@ class GraphSAGE(nn.Module):
             x_target = x[:size[1]]
             x = self.convs[i]((x, x_target), edge_index, size)
             if i != self.num_layers - 1:
-                x = F.relu(x)
-                x = F.dropout(x, p=self.dropout, training=self.training)
+                x = F.relu(x, inplace=True)
+                if self.dropout > 0:
+                    x = F.dropout(x, p=self.dropout, training=self.training)
         return torch.log_softmax(x, dim=-1)
 
     def full_forward(self, x, edge_index):
@@ -272,6 +285,7 @@ class GAT(nn.Module):
             x_target = x[:size[1]]
             x = self.convs[i]((x, x_target), edge_index, size)
             if i != self.num_layers - 1:
-                x = F.elu(x)
-                x = F.dropout(x, p=self.dropout, training=self.training)
+                x = F.elu(x, inplace=True)
+                if self.dropout > 0:
+                    x = F.dropout(x, p=self.dropout, training=self.training)
         return torch.log_softmax(x, dim=-1)
