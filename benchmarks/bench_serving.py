@@ -184,8 +184,16 @@ def main():
                        daemon=True)
         c.start()
         clients.append(c)
+    deadline = time.time() + args.seconds + 90
+    stalled = False
     for c in clients:
-        c.join()
+        c.join(timeout=max(1.0, deadline - time.time()))
+        if c.is_alive():
+            stalled = True
+    if stalled:
+        print("# WARNING: client(s) still blocked flushing the stream "
+              "queue after the deadline — pipeline cannot absorb the "
+              "offered load; dumping stacks and continuing", flush=True)
     batcher.stop()
 
     def dump_worker_stacks():
@@ -203,6 +211,9 @@ def main():
             except (OSError, TypeError):
                 pass
         time.sleep(3)
+
+    if stalled:
+        dump_worker_stacks()
 
     stats = []
     deadline = time.time() + 120

@@ -16,6 +16,7 @@
 
 #include <algorithm>
 #include <atomic>
+#include <thread>
 #include <random>
 #include <unordered_map>
 #include <vector>
@@ -51,6 +52,27 @@ struct DeviceScope {
     }
     ~DeviceScope() { (void)hipSetDevice(prev); }
 };
+
+// Plain std::thread fan-out over chunk ranges.  at::parallel_for called
+// from a non-main thread (the prefetch worker) degrades to serial
+// execution, which made the CPU-staged gather slower than the zero-copy
+// kernel it replaces; explicit threads sidestep the intra-op pool.
+template <typename F>
+void parallel_chunks(int64_t nchunk, int nthreads, F&& fn) {
+    if (nthreads > nchunk) nthreads = (int)nchunk;
+    if (nthreads <= 1) {
+        fn((int64_t)0, nchunk);
+        return;
+    }
+    std::vector<std::thread> ts;
+    int64_t per = (nchunk + nthreads - 1) / nthreads;
+    for (int t = 0; t < nthreads; ++t) {
+        int64_t b = (int64_t)t * per, e = std::min<int64_t>(nchunk, b + per);
+        if (b >= e) break;
+        ts.emplace_back([&fn, b, e] { fn(b, e); });
+    }
+    for (auto& th : ts) th.join();
+}
 
 // Exclusive scan of an int64 device tensor; returns (prefix, total-on-host).
 std::pair<torch::Tensor, int64_t> exclusive_scan_total(
@@ -735,7 +757,7 @@ class ShardTensor {
                 if (v >= h.beg && v < h.end) return &h;
             return nullptr;
         };
-        at::parallel_for(0, nchunk, 1, [&](int64_t c0, int64_t c1) {
+        parallel_chunks(nchunk, 32, [&](int64_t c0, int64_t c1) {
             for (int64_t c = c0; c < c1; ++c) {
                 int64_t cnt = 0;
                 int64_t e = std::min(n, (c + 1) * grain);
@@ -755,7 +777,7 @@ class ShardTensor {
             stage_pin_ = torch::empty({std::max(m * rb, n * rb)}, pin_opts);
         int64_t* pos = (int64_t*)pos_pin_.data_ptr();
         char* stage = (char*)stage_pin_.data_ptr();
-        at::parallel_for(0, nchunk, 1, [&](int64_t c0, int64_t c1) {
+        parallel_chunks(nchunk, 64, [&](int64_t c0, int64_t c1) {
             for (int64_t c = c0; c < c1; ++c) {
                 int64_t j = chunk_off[c];
                 int64_t e = std::min(n, (c + 1) * grain);
