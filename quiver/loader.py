@@ -1,11 +1,15 @@
 """Prefetching sample+gather pipeline.
 
-Overlaps hop sampling and feature gathering of batch i+1 with model
-compute of batch i: a worker thread drives the sampler and feature store
-on its own HIP stream (the native calls release the GIL, so the main
-thread keeps launching model kernels).  The reference runs these stages
-sequentially; on MI355X the sampler's small kernels and the gather's
-PCIe-bound cold reads coexist well with the model's GEMMs.
+Overlaps hop sampling and feature gathering of batch i+1..i+depth with
+model compute of batch i — WITHOUT a worker thread.  A two-thread design
+(v1) serialized in practice: the producer thread was GIL-starved by the
+main thread's kernel-launch loop, so sampling for batch i+1 only started
+after compute for batch i had finished (kernel-trace evidence in
+profiles/).  Here the main thread drives everything on one side HIP
+stream: right after the training step for batch i is LAUNCHED (async),
+it produces batch i+depth — the one host sync inside the fused sampler
+(`sample_hops` sizes readback) blocks the CPU while the GPU concurrently
+executes compute i on the main stream and sampling on the side stream.
 
 Usage:
     loader = quiver.TrainingPrefetcher(sampler, feature, batches, depth=2)
@@ -13,16 +17,11 @@ Usage:
         out = model(x, adjs)
         ...
 """
-import queue
-import threading
+from collections import deque
 
 import torch
 
 __all__ = ["TrainingPrefetcher"]
-
-
-class _End:
-    pass
 
 
 class TrainingPrefetcher:
@@ -30,59 +29,45 @@ class TrainingPrefetcher:
         self.sampler = sampler
         self.feature = feature
         self.seed_batches = seed_batches
-        self.depth = depth
+        self.depth = max(1, depth)
         self.device = device if device is not None \
             else torch.cuda.current_device()
 
     def __iter__(self):
-        q = queue.Queue(maxsize=self.depth)
-        stop = threading.Event()
-        stream = torch.cuda.Stream(self.device)
+        cur = torch.cuda.current_stream(self.device)
+        side = torch.cuda.Stream(self.device)
+        it = iter(self.seed_batches)
+        pending = deque()
 
-        def worker():
+        def produce():
             try:
-                with torch.cuda.stream(stream):
-                    for seeds in self.seed_batches:
-                        if stop.is_set():
-                            break
-                        n_id, bs, adjs = self.sampler.sample(seeds)
-                        x = self.feature[n_id] if self.feature is not None \
-                            else None
-                        ev = torch.cuda.Event()
-                        ev.record(stream)
-                        q.put((n_id, bs, adjs, x, ev))
-            except BaseException as e:  # propagate to consumer
-                q.put(e)
-                return
-            q.put(_End())
+                seeds = next(it)
+            except StopIteration:
+                return False
+            with torch.cuda.stream(side):
+                n_id, bs, adjs = self.sampler.sample(seeds)
+                x = self.feature[n_id] if self.feature is not None else None
+                ev = torch.cuda.Event()
+                ev.record(side)
+            pending.append((n_id, bs, adjs, x, ev))
+            return True
 
-        t = threading.Thread(target=worker, daemon=True)
-        t.start()
-        try:
-            cur = torch.cuda.current_stream(self.device)
-            while True:
-                item = q.get()
-                if isinstance(item, _End):
-                    break
-                if isinstance(item, BaseException):
-                    raise item
-                n_id, bs, adjs, x, ev = item
-                # main stream waits for the side stream's work...
-                cur.wait_event(ev)
-                # ...and the producer-stream allocations must not be reused
-                # until main-stream work on them completes
-                n_id.record_stream(cur)
-                for adj in adjs:
-                    adj.edge_index.record_stream(cur)
-                if x is not None:
-                    x.record_stream(cur)
-                yield n_id, bs, adjs, x
-        finally:
-            stop.set()
-            # drain so the worker can exit
-            try:
-                while True:
-                    q.get_nowait()
-            except queue.Empty:
-                pass
-            t.join(timeout=10)
+        for _ in range(self.depth):
+            if not produce():
+                break
+        while pending:
+            n_id, bs, adjs, x, ev = pending.popleft()
+            # main stream waits for the side stream's work for THIS batch...
+            cur.wait_event(ev)
+            # ...and side-stream allocations must not be reused until
+            # main-stream work on them completes
+            n_id.record_stream(cur)
+            for adj in adjs:
+                adj.edge_index.record_stream(cur)
+            if x is not None:
+                x.record_stream(cur)
+            # consumer launches the training step for this batch inside the
+            # yield; on re-entry we produce the next batch so its sample
+            # sync overlaps that compute
+            yield n_id, bs, adjs, x
+            produce()
