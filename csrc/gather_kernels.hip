@@ -25,7 +25,9 @@ struct alignas(16) vec16 { uint64_t a, b; };
 template <typename VecT, int SUB, bool SCATTER>
 __global__ void __launch_bounds__(BLOCK)
 copy_rows_kernel(GatherSpec spec, const int64_t* __restrict__ indices,
-                 int64_t n, char* __restrict__ other) {
+                 int64_t n, char* __restrict__ other,
+                 const int64_t* __restrict__ n_dev) {
+    if (n_dev) n = min(n, *n_dev);
     const int sub_id = threadIdx.x / SUB;
     const int lane = threadIdx.x % SUB;
     const int rows_per_block = BLOCK / SUB;
@@ -79,7 +81,8 @@ inline int gather_max_blocks(const GatherSpec& spec) {
 
 template <typename VecT, bool SCATTER>
 void dispatch_sub(hipStream_t s, const GatherSpec& spec,
-                  const int64_t* indices, int64_t n, char* other) {
+                  const int64_t* indices, int64_t n, char* other,
+                  const int64_t* n_dev) {
     int64_t nvec = spec.row_bytes / (int64_t)sizeof(VecT);
     int sub = 4;
     while (sub < 64 && sub < nvec) sub *= 2;  // cover the row in ~1 pass
@@ -88,7 +91,7 @@ void dispatch_sub(hipStream_t s, const GatherSpec& spec,
 #define QK_CASE(S)                                                          \
     case S:                                                                 \
         copy_rows_kernel<VecT, S, SCATTER>                                  \
-            <<<grid, BLOCK, 0, s>>>(spec, indices, n, other);               \
+            <<<grid, BLOCK, 0, s>>>(spec, indices, n, other, n_dev);        \
         break;
         QK_CASE(4) QK_CASE(8) QK_CASE(16) QK_CASE(32) QK_CASE(64)
 #undef QK_CASE
@@ -98,28 +101,29 @@ void dispatch_sub(hipStream_t s, const GatherSpec& spec,
 
 template <bool SCATTER>
 void copy_rows(hipStream_t s, const GatherSpec& spec, const int64_t* indices,
-               int64_t n, char* other) {
+               int64_t n, char* other, const int64_t* n_dev) {
     if (n == 0 || spec.nshards == 0) return;
     if (spec.row_bytes % 16 == 0)
-        dispatch_sub<vec16, SCATTER>(s, spec, indices, n, other);
+        dispatch_sub<vec16, SCATTER>(s, spec, indices, n, other, n_dev);
     else if (spec.row_bytes % 8 == 0)
-        dispatch_sub<uint64_t, SCATTER>(s, spec, indices, n, other);
+        dispatch_sub<uint64_t, SCATTER>(s, spec, indices, n, other, n_dev);
     else if (spec.row_bytes % 4 == 0)
-        dispatch_sub<uint32_t, SCATTER>(s, spec, indices, n, other);
+        dispatch_sub<uint32_t, SCATTER>(s, spec, indices, n, other, n_dev);
     else
-        dispatch_sub<uint8_t, SCATTER>(s, spec, indices, n, other);
+        dispatch_sub<uint8_t, SCATTER>(s, spec, indices, n, other, n_dev);
 }
 
 }  // namespace
 
 void launch_gather(hipStream_t s, const GatherSpec& spec,
-                   const int64_t* indices, int64_t n, char* out) {
-    copy_rows<false>(s, spec, indices, n, out);
+                   const int64_t* indices, int64_t n, char* out,
+                   const int64_t* n_dev) {
+    copy_rows<false>(s, spec, indices, n, out, n_dev);
 }
 
 void launch_scatter(hipStream_t s, const GatherSpec& spec,
                     const int64_t* indices, int64_t n, const char* src) {
-    copy_rows<true>(s, spec, indices, n, const_cast<char*>(src));
+    copy_rows<true>(s, spec, indices, n, const_cast<char*>(src), nullptr);
 }
 
 }  // namespace qk

@@ -223,8 +223,17 @@ class GpuSampler {
     // kernel-bound (profiles/: both streams <50% busy).
     // Returns per hop: (frontier, row_idx, col_idx) with frontier[0:n_prev]
     // == previous hop's frontier (seeds first), sizes already exact.
-    std::vector<std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>>
-    sample_hops(torch::Tensor seeds, const std::vector<int>& ks) {
+    // Raw variant: returns UPPER-BOUND-sized per-hop tensors plus the
+    // device-resident sizes vector [m_0, u_0, m_1, u_1, ...] WITHOUT any
+    // host synchronization — callers chain the feature gather behind it
+    // (gather kernel reads the exact frontier count from sizes_dev) and
+    // read the sizes only when the whole chain has been consumed.
+    // Frontier slack is zeroed so downstream indexing (feature_order
+    // remap, gather) stays in-range.
+    std::tuple<
+        std::vector<std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>>,
+        torch::Tensor>
+    sample_hops_raw(torch::Tensor seeds, const std::vector<int>& ks) {
         DeviceScope g(device_);
         auto stream = current_stream();
         seeds = seeds.contiguous();
@@ -294,7 +303,7 @@ class GpuSampler {
                                       flags.data_ptr<int64_t>(),
                                       scanned.data_ptr<int64_t>(), total_ub,
                                       sd + 2 * h + 1);  // unique_h
-            auto frontier = torch::empty({total_ub}, opts);
+            auto frontier = torch::zeros({total_ub}, opts);
             qk::launch_compact_unique(
                 stream, keys.data_ptr<int64_t>(), local.data_ptr<int32_t>(),
                 pos.data_ptr<int32_t>(), capacity, cur.data_ptr<int64_t>(),
@@ -320,16 +329,26 @@ class GpuSampler {
             n_ub = total_ub;
         }
 
+        std::vector<std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>>
+            raw;
+        for (int h = 0; h < H; ++h)
+            raw.emplace_back(frontiers[h], rows[h], cols[h]);
+        return {raw, sizes_dev};
+    }
+
+    std::vector<std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>>
+    sample_hops(torch::Tensor seeds, const std::vector<int>& ks) {
+        auto [raw, sizes_dev] = sample_hops_raw(seeds, ks);
         // the ONE sync of the whole batch
         auto sizes_host = sizes_dev.cpu();
         const int64_t* sh = sizes_host.data_ptr<int64_t>();
         std::vector<std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>>
             res;
-        for (int h = 0; h < H; ++h) {
+        for (size_t h = 0; h < raw.size(); ++h) {
             int64_t m = sh[2 * h], u = sh[2 * h + 1];
-            res.emplace_back(frontiers[h].narrow(0, 0, u),
-                             rows[h].narrow(0, 0, m),
-                             cols[h].narrow(0, 0, m));
+            res.emplace_back(std::get<0>(raw[h]).narrow(0, 0, u),
+                             std::get<1>(raw[h]).narrow(0, 0, m),
+                             std::get<2>(raw[h]).narrow(0, 0, m));
         }
         return res;
     }
@@ -623,6 +642,17 @@ class ShardTensor {
     // cross-clique fallback: remote device reads its own HBM, result is
     // copied back by the python layer).
     torch::Tensor gather_on(int dev, torch::Tensor indices) {
+        return gather_impl(dev, indices, torch::Tensor());
+    }
+
+    // n_dev: 1-element int64 CUDA tensor holding the exact row count
+    // (indices is upper-bound sized); enables fully async chains.
+    torch::Tensor gather_n(torch::Tensor indices, torch::Tensor n_dev) {
+        return gather_impl(device_, indices, n_dev);
+    }
+
+    torch::Tensor gather_impl(int dev, torch::Tensor indices,
+                              torch::Tensor n_dev_t) {
         DeviceScope g(dev);
         indices = indices.contiguous();
         TORCH_CHECK(indices.device().is_cuda(), "indices must be on GPU");
@@ -634,6 +664,9 @@ class ShardTensor {
                        torch::Device(torch::kCUDA, dev)));
         auto spec = build_spec(dev);
         auto cur = c10::hip::getCurrentHIPStream(dev);
+        const int64_t* n_dev = n_dev_t.defined()
+                                   ? n_dev_t.data_ptr<int64_t>()
+                                   : nullptr;
 
         // Tier-split: a mixed gather would run at the capped host grid for
         // ALL rows (the cap keeps the PCIe-latency-bound zero-copy pass
@@ -646,7 +679,7 @@ class ShardTensor {
         host_mask &= spec.access_mask;
         uint32_t dev_mask = spec.access_mask & ~host_mask;
 
-        if (host_mask != 0 && dev_mask != 0 && staged_wanted(n)) {
+        if (host_mask != 0 && dev_mask != 0 && !n_dev && staged_wanted(n)) {
             // CPU-staged host tier: zero-copy kernel reads of scattered
             // 400 B rows top out near ~11 GB/s (64 B uncached PCIe reads,
             // latency-bound).  Instead: HBM/xGMI rows via the full-grid
@@ -679,9 +712,9 @@ class ShardTensor {
             QK_CHECK_HIP(hipEventRecord(ev.first, cur.stream()));
             QK_CHECK_HIP(hipStreamWaitEvent(side.stream(), ev.first, 0));
             qk::launch_gather(cur.stream(), hs, indices.data_ptr<int64_t>(),
-                              n, (char*)out.data_ptr());
+                              n, (char*)out.data_ptr(), n_dev);
             qk::launch_gather(side.stream(), ds, indices.data_ptr<int64_t>(),
-                              n, (char*)out.data_ptr());
+                              n, (char*)out.data_ptr(), n_dev);
             QK_CHECK_HIP(hipEventRecord(ev.second, side.stream()));
             QK_CHECK_HIP(hipStreamWaitEvent(cur.stream(), ev.second, 0));
             // caching-allocator hazard: out/indices are used on `side`.
@@ -694,7 +727,7 @@ class ShardTensor {
         } else {
             qk::launch_gather(cur.stream(), spec,
                               indices.data_ptr<int64_t>(), n,
-                              (char*)out.data_ptr());
+                              (char*)out.data_ptr(), n_dev);
         }
         return out;
     }
@@ -1090,6 +1123,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         .def("sample_hops", &GpuSampler::sample_hops,
              py::call_guard<py::gil_scoped_release>(),
              "fused multi-hop sample+reindex, one sync per batch")
+        .def("sample_hops_raw", &GpuSampler::sample_hops_raw,
+             py::call_guard<py::gil_scoped_release>(),
+             "zero-sync variant: ub-sized tensors + device sizes vector")
         .def("reindex_single", &GpuSampler::reindex_single,
              py::call_guard<py::gil_scoped_release>())
         .def("cal_neighbor_prob", &GpuSampler::cal_neighbor_prob,
@@ -1136,6 +1172,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         .def("gather", &ShardTensor::gather,
              py::call_guard<py::gil_scoped_release>())
         .def("gather_on", &ShardTensor::gather_on,
+             py::call_guard<py::gil_scoped_release>())
+        .def("gather_n", &ShardTensor::gather_n,
              py::call_guard<py::gil_scoped_release>())
         .def("access_mask_on", &ShardTensor::access_mask_on)
         .def("shard_ends", &ShardTensor::shard_ends)

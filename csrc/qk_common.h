@@ -122,8 +122,11 @@ struct GatherSpec {
 
 // out[i] = row indices[i] of the virtual tensor; rows whose shard is not
 // accessible are left untouched (python layer fills them via a peer pass).
+// n is a host upper bound; when n_dev != nullptr the exact row count is
+// read from the device (async sample+gather chains, no host sync).
 void launch_gather(hipStream_t s, const GatherSpec& spec,
-                   const int64_t* indices, int64_t n, char* out);
+                   const int64_t* indices, int64_t n, char* out,
+                   const int64_t* n_dev = nullptr);
 
 // Scatter-style update used by the python layer for cache fill / tests:
 // shard-resident rows only.  dst row indices[i] <- src[i].

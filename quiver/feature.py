@@ -212,6 +212,20 @@ class Feature(object):
             return self.device_tensor_list[self.rank]
         return self.clique_tensor_list[self.topo.get_clique_id(self.rank)]
 
+    def gather_raw(self, node_idx, n_dev):
+        """Async gather over an upper-bound-sized frontier: the exact row
+        count is read by the kernel from the 1-element device tensor
+        `n_dev` — no host synchronization anywhere on the path.  Slack
+        output rows are untouched; the caller slices.  Not available for
+        the disk (mmap) tier or partially-accessible shard layouts."""
+        self.lazy_init_from_ipc_handle()
+        if self.mmap_handle_ is not None:
+            raise RuntimeError("gather_raw does not support the disk tier")
+        node_idx = node_idx.to(self.rank)
+        if self.feature_order is not None:
+            node_idx = self.feature_order[node_idx]
+        return self._shard_tensor().gather_n(node_idx, n_dev)
+
     def __getitem__(self, node_idx: torch.Tensor):
         self.lazy_init_from_ipc_handle()
         node_idx = node_idx.to(self.rank)

@@ -33,11 +33,28 @@ class TrainingPrefetcher:
         self.device = device if device is not None \
             else torch.cuda.current_device()
 
+    def _can_chain_async(self):
+        """Zero-sync produce path: fused GPU sampler + directly-accessible
+        feature shards (no disk tier, no cross-clique pass)."""
+        if not hasattr(self.sampler, "sample_async"):
+            return False
+        if getattr(self.sampler, "mode", None) not in ("GPU", "UVA"):
+            return False
+        if getattr(self.sampler, "sort_frontier", False):
+            return False
+        if self.feature is not None:
+            if not hasattr(self.feature, "gather_raw"):
+                return False
+            if getattr(self.feature, "mmap_handle_", None) is not None:
+                return False
+        return True
+
     def __iter__(self):
         cur = torch.cuda.current_stream(self.device)
         side = torch.cuda.Stream(self.device)
         it = iter(self.seed_batches)
         pending = deque()
+        chain_async = [self._can_chain_async()]
 
         def produce():
             try:
@@ -45,29 +62,57 @@ class TrainingPrefetcher:
             except StopIteration:
                 return False
             with torch.cuda.stream(side):
+                if chain_async[0]:
+                    try:
+                        tok = self.sampler.sample_async(seeds)
+                        x_ub = None
+                        if self.feature is not None:
+                            raw, sizes_dev = tok[1], tok[2]
+                            n_dev = sizes_dev[2 * len(raw) - 1:2 * len(raw)]
+                            x_ub = self.feature.gather_raw(raw[-1][0], n_dev)
+                        ev = torch.cuda.Event()
+                        ev.record(side)
+                        pending.append(("tok", tok, x_ub, ev))
+                        return True
+                    except RuntimeError:
+                        chain_async[0] = False  # fall through, stay sync
                 n_id, bs, adjs = self.sampler.sample(seeds)
                 x = self.feature[n_id] if self.feature is not None else None
                 ev = torch.cuda.Event()
                 ev.record(side)
-            pending.append((n_id, bs, adjs, x, ev))
+            pending.append(("done", (n_id, bs, adjs), x, ev))
             return True
 
         for _ in range(self.depth):
             if not produce():
                 break
         while pending:
-            n_id, bs, adjs, x, ev = pending.popleft()
-            # main stream waits for the side stream's work for THIS batch...
+            kind, payload, x, ev = pending.popleft()
+            # main stream waits for the side stream's work for THIS batch
             cur.wait_event(ev)
-            # ...and side-stream allocations must not be reused until
-            # main-stream work on them completes
+            if kind == "tok":
+                # CPU-side wait for THIS batch only (event, not stream):
+                # makes the pinned sizes valid; later batches keep running
+                ev.synchronize()
+                n_id, bs, adjs = self.sampler.sample_finalize(payload)
+                # finalize's torch.stack reads the side-allocated raw
+                # buffers on the main stream
+                for f_ub, row_ub, col_ub in payload[1]:
+                    row_ub.record_stream(cur)
+                    col_ub.record_stream(cur)
+                if x is not None:
+                    x = x[:n_id.size(0)]
+            else:
+                n_id, bs, adjs = payload
+            # side-stream allocations must not be reused until main-stream
+            # work on them completes
             n_id.record_stream(cur)
             for adj in adjs:
                 adj.edge_index.record_stream(cur)
             if x is not None:
                 x.record_stream(cur)
             # consumer launches the training step for this batch inside the
-            # yield; on re-entry we produce the next batch so its sample
-            # sync overlaps that compute
+            # yield; on re-entry we produce batch i+depth so the whole
+            # sample+gather chain overlaps that compute
             yield n_id, bs, adjs, x
             produce()

@@ -149,6 +149,49 @@ class GraphSageSampler:
             nodes = frontier
         return nodes, batch_size, adjs[::-1]
 
+    def sample_async(self, input_nodes):
+        """Enqueue one full multi-hop sample with ZERO host syncs.
+
+        Returns a token for :meth:`sample_finalize`.  The per-hop tensors
+        in the token are upper-bound sized; exact sizes land in a pinned
+        host buffer via an async D2H, valid once the enqueueing stream's
+        work for this batch completes (callers order with an event).
+        GPU/UVA modes with positive fanouts only.
+        """
+        self.lazy_init_quiver()
+        if self.mode not in ("GPU", "UVA") or not all(
+                s > 0 for s in self.sizes):
+            raise RuntimeError("sample_async requires GPU/UVA mode with "
+                               "positive fanouts")
+        if not isinstance(input_nodes, torch.Tensor):
+            input_nodes = torch.tensor(input_nodes, dtype=torch.long)
+        nodes = input_nodes.to(self.device)
+        raw, sizes_dev = self.quiver.sample_hops_raw(nodes, self.sizes)
+        sizes_pin = torch.empty(sizes_dev.numel(), dtype=torch.int64,
+                                pin_memory=True)
+        sizes_pin.copy_(sizes_dev, non_blocking=True)
+        return (nodes, raw, sizes_dev, sizes_pin)
+
+    def sample_finalize(self, token):
+        """Build (n_id, batch_size, adjs) from a sample_async token.
+
+        The caller must have synchronized with the producing stream (e.g.
+        event.synchronize()) so the pinned sizes are valid.
+        """
+        nodes, raw, _sizes_dev, sizes_pin = token
+        adjs = []
+        prev = nodes.size(0)
+        frontier = nodes
+        for h, (f_ub, row_ub, col_ub) in enumerate(raw):
+            m = int(sizes_pin[2 * h])
+            u = int(sizes_pin[2 * h + 1])
+            frontier = f_ub[:u]
+            edge_index = torch.stack([col_ub[:m], row_ub[:m]], dim=0)
+            adjs.append(Adj(edge_index, torch.tensor([]),
+                            torch.LongTensor([u, prev])))
+            prev = u
+        return frontier, nodes.size(0), adjs[::-1]
+
     def sample_prob(self, train_idx, total_node_count):
         """Multi-hop access probability of every node when seeding from
         train_idx — drives access-probability feature placement."""

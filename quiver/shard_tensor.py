@@ -117,6 +117,15 @@ class ShardTensor:
         result = self.shard_tensor.gather_on(inter_device, request_nodes)
         wait_results.append((part_orders, result.to(self.current_device)))
 
+    def gather_n(self, nodes, n_dev):
+        """Upper-bound gather with device-side exact count (async path);
+        requires every shard to be directly accessible from this device."""
+        if self._inaccessible_ranges():
+            raise RuntimeError("gather_n: some shards need the cross-clique "
+                               "pass; use __getitem__")
+        nodes = nodes.to(self.current_device)
+        return self.shard_tensor.gather_n(nodes, n_dev)
+
     def __getitem__(self, nodes):
         nodes = nodes.to(self.current_device)
         feature = self.shard_tensor[nodes]
