@@ -74,3 +74,27 @@ def test_topo_single_clique():
     topo = quiver.p2pCliqueTopo([0])
     assert topo.p2p_clique_count == 1
     assert topo.get_clique_id(0) == 0
+
+
+def test_prefetcher_async_chain_correct(training_setup):
+    """TrainingPrefetcher's zero-sync chain yields the same feature rows a
+    direct (synchronous) gather of the yielded frontier produces."""
+    topo, x, y = training_setup
+    sampler = quiver.GraphSageSampler(topo, [10, 5], device=0, mode="GPU")
+    feature = quiver.Feature(0, device_list=[0], device_cache_size="1M",
+                             cache_policy="device_replicate", csr_topo=topo)
+    feature.from_cpu_tensor(x)
+    batches = [torch.randint(0, topo.node_count, (256,))
+               for _ in range(6)]
+    seen = 0
+    for n_id, bs, adjs, feats in quiver.TrainingPrefetcher(
+            sampler, feature, batches, depth=2, device=0):
+        assert bs == 256
+        assert feats.shape[0] == n_id.shape[0]
+        ref = feature[n_id]
+        assert torch.equal(feats, ref)
+        for adj in adjs:
+            assert adj.edge_index[0].max() < adj.size[0]
+            assert adj.edge_index[1].max() < adj.size[1]
+        seen += 1
+    assert seen == len(batches)
