@@ -51,7 +51,9 @@ class TrainingPrefetcher:
 
     def __iter__(self):
         cur = torch.cuda.current_stream(self.device)
-        side = torch.cuda.Stream(self.device)
+        # high-priority stream: the PCIe-latency-bound host gather needs its
+        # blocks resident promptly even while model kernels churn CU slots
+        side = torch.cuda.Stream(self.device, priority=-1)
         it = iter(self.seed_batches)
         pending = deque()
         chain_async = [self._can_chain_async()]
