@@ -90,6 +90,7 @@ def main():
     p.add_argument("--edges", type=int, default=None)
     p.add_argument("--no-overlap", action="store_true",
                    help="disable the sample+gather / compute prefetch overlap")
+    p.add_argument("--prefetch-streams", type=int, default=1)
     args = p.parse_args()
 
     ps = PRESETS[args.preset]
@@ -193,7 +194,8 @@ def main():
         else:
             pf = quiver.TrainingPrefetcher(sampler, feature,
                                            batches[lo:hi], depth=2,
-                                           device=local_rank)
+                                           device=local_rank,
+                                           num_streams=args.prefetch_streams)
             for n_id, bs, adjs, x in pf:
                 total += train_on(n_id, bs, adjs, x)
         return total

@@ -25,11 +25,13 @@ __all__ = ["TrainingPrefetcher"]
 
 
 class TrainingPrefetcher:
-    def __init__(self, sampler, feature, seed_batches, depth=2, device=None):
+    def __init__(self, sampler, feature, seed_batches, depth=2, device=None,
+                 num_streams=1):
         self.sampler = sampler
         self.feature = feature
         self.seed_batches = seed_batches
-        self.depth = max(1, depth)
+        self.depth = max(1, depth, num_streams)
+        self.num_streams = max(1, num_streams)
         self.device = device if device is not None \
             else torch.cuda.current_device()
 
@@ -51,18 +53,24 @@ class TrainingPrefetcher:
 
     def __iter__(self):
         cur = torch.cuda.current_stream(self.device)
-        # high-priority stream: the PCIe-latency-bound host gather needs its
-        # blocks resident promptly even while model kernels churn CU slots
-        side = torch.cuda.Stream(self.device, priority=-1)
+        # high-priority streams: the PCIe-latency-bound host gather needs
+        # its blocks resident promptly even while model kernels churn CU
+        # slots.  num_streams=2 round-robins batches over two streams so
+        # one batch's sample phase hides under the other's gather.
+        sides = [torch.cuda.Stream(self.device, priority=-1)
+                 for _ in range(self.num_streams)]
         it = iter(self.seed_batches)
         pending = deque()
         chain_async = [self._can_chain_async()]
+        rr = [0]
 
         def produce():
             try:
                 seeds = next(it)
             except StopIteration:
                 return False
+            side = sides[rr[0] % len(sides)]
+            rr[0] += 1
             with torch.cuda.stream(side):
                 if chain_async[0]:
                     try:
