@@ -16,6 +16,9 @@
 
 #include <algorithm>
 #include <atomic>
+#include <condition_variable>
+#include <functional>
+#include <mutex>
 #include <thread>
 #include <random>
 #include <unordered_map>
@@ -53,25 +56,84 @@ struct DeviceScope {
     ~DeviceScope() { (void)hipSetDevice(prev); }
 };
 
-// Plain std::thread fan-out over chunk ranges.  at::parallel_for called
-// from a non-main thread (the prefetch worker) degrades to serial
-// execution, which made the CPU-staged gather slower than the zero-copy
-// kernel it replaces; explicit threads sidestep the intra-op pool.
+// Persistent worker pool for the CPU-staged gather.  Spawning threads
+// per call costs ~2 ms (64 x ~30 us) — more than the copy itself — and
+// at::parallel_for from a non-main thread degrades to serial; a resident
+// pool with an atomic chunk cursor avoids both.
+class WorkerPool {
+  public:
+    static WorkerPool& inst() {
+        // leaked singleton: worker threads must not race static
+        // destruction at process exit
+        static WorkerPool* p = new WorkerPool();
+        return *p;
+    }
+
+    template <typename F>
+    void run(int64_t nchunk, F&& fn) {
+        if (nchunk <= 1) {
+            if (nchunk == 1) fn((int64_t)0);
+            return;
+        }
+        {
+            std::unique_lock<std::mutex> lk(mu_);
+            job_ = [&fn](int64_t c) { fn(c); };
+            nchunk_ = nchunk;
+            cursor_.store(0);
+            pending_.store((int)workers_.size());
+            ++epoch_;
+        }
+        cv_.notify_all();
+        // the caller helps
+        drain();
+        std::unique_lock<std::mutex> lk(mu_);
+        done_cv_.wait(lk, [this] { return pending_.load() == 0; });
+        job_ = nullptr;
+    }
+
+  private:
+    WorkerPool() {
+        unsigned hw = std::thread::hardware_concurrency();
+        int n = std::min(63u, hw > 2 ? hw - 2 : 1u);
+        for (int i = 0; i < n; ++i)
+            workers_.emplace_back([this] { worker(); });
+    }
+    void drain() {
+        int64_t c;
+        while ((c = cursor_.fetch_add(1)) < nchunk_) job_(c);
+    }
+    void worker() {
+        uint64_t seen = 0;
+        while (true) {
+            std::unique_lock<std::mutex> lk(mu_);
+            cv_.wait(lk, [&] { return epoch_ != seen; });
+            seen = epoch_;
+            auto job = job_;
+            int64_t total = nchunk_;
+            lk.unlock();
+            if (job) {
+                int64_t c;
+                while ((c = cursor_.fetch_add(1)) < total) job(c);
+            }
+            if (pending_.fetch_sub(1) == 1) {
+                std::lock_guard<std::mutex> g(mu_);
+                done_cv_.notify_all();
+            }
+        }
+    }
+    std::vector<std::thread> workers_;
+    std::mutex mu_;
+    std::condition_variable cv_, done_cv_;
+    std::function<void(int64_t)> job_;
+    std::atomic<int64_t> cursor_{0};
+    int64_t nchunk_ = 0;
+    std::atomic<int> pending_{0};
+    uint64_t epoch_ = 0;
+};
+
 template <typename F>
-void parallel_chunks(int64_t nchunk, int nthreads, F&& fn) {
-    if (nthreads > nchunk) nthreads = (int)nchunk;
-    if (nthreads <= 1) {
-        fn((int64_t)0, nchunk);
-        return;
-    }
-    std::vector<std::thread> ts;
-    int64_t per = (nchunk + nthreads - 1) / nthreads;
-    for (int t = 0; t < nthreads; ++t) {
-        int64_t b = (int64_t)t * per, e = std::min<int64_t>(nchunk, b + per);
-        if (b >= e) break;
-        ts.emplace_back([&fn, b, e] { fn(b, e); });
-    }
-    for (auto& th : ts) th.join();
+void parallel_chunks(int64_t nchunk, int /*nthreads*/, F&& fn) {
+    WorkerPool::inst().run(nchunk, [&fn](int64_t c) { fn(c, c + 1); });
 }
 
 // Exclusive scan of an int64 device tensor; returns (prefix, total-on-host).
@@ -747,12 +809,13 @@ class ShardTensor {
             const char* e = getenv("QUIVER_STAGED_GATHER");
             return e ? atoi(e) : -1;  // -1 = auto
         }();
+        if (mode == 0) return false;
         if (mode == 1) return true;
-        // default OFF: measured 19.6 GB/s vs 71 GB/s for the zero-copy
-        // kernel on the products shape (CPU random-row reads are the
-        // bottleneck, not PCIe) — kept as an opt-in experiment
-        (void)n;
-        return false;
+        // auto: large training gathers.  The zero-copy kernel's uncached
+        // PCIe reads slow CONCURRENT kernels 5-10x (fabric poisoning,
+        // profiles/timeline_analysis.md), so overlap never pays; the
+        // staged path keeps host traffic on the SDMA engines.
+        return n >= 16384;
     }
 
     // CPU-staged gather of the pinned-host tier (see gather_on).
