@@ -741,7 +741,7 @@ class ShardTensor {
         host_mask &= spec.access_mask;
         uint32_t dev_mask = spec.access_mask & ~host_mask;
 
-        if (host_mask != 0 && dev_mask != 0 && !n_dev && staged_wanted(n)) {
+        if (host_mask != 0 && dev_mask != 0 && staged_wanted(n)) {
             // CPU-staged host tier: zero-copy kernel reads of scattered
             // 400 B rows top out near ~11 GB/s (64 B uncached PCIe reads,
             // latency-bound).  Instead: HBM/xGMI rows via the full-grid
@@ -753,9 +753,9 @@ class ShardTensor {
             ds.access_mask = dev_mask;
             ds.has_host_shard = false;
             qk::launch_gather(cur.stream(), ds, indices.data_ptr<int64_t>(),
-                              n, (char*)out.data_ptr());
+                              n, (char*)out.data_ptr(), n_dev);
             gather_host_staged(dev, cur.stream(), indices, n, spec,
-                               host_mask, out);
+                               host_mask, out, n_dev);
         } else if (host_mask != 0 && dev_mask != 0) {
             qk::GatherSpec hs = spec;
             hs.access_mask = host_mask;
@@ -821,7 +821,8 @@ class ShardTensor {
     // CPU-staged gather of the pinned-host tier (see gather_on).
     void gather_host_staged(int dev, hipStream_t stream, torch::Tensor indices,
                             int64_t n, const qk::GatherSpec& spec,
-                            uint32_t host_mask, torch::Tensor& out) {
+                            uint32_t host_mask, torch::Tensor& out,
+                            const int64_t* n_dev = nullptr) {
         const int64_t rb = row_bytes_;
         // host base pointer + global row range of every host shard
         struct HostShard { int64_t beg, end; const char* base; };
@@ -838,13 +839,19 @@ class ShardTensor {
                             .dtype(torch::kUInt8)
                             .device(torch::kCPU)
                             .pinned_memory(true);
-        if (!idx_pin_.defined() || idx_pin_.numel() < n * 8)
-            idx_pin_ = torch::empty({n * 8}, pin_opts);
+        if (!idx_pin_.defined() || idx_pin_.numel() < n * 8 + 8)
+            idx_pin_ = torch::empty({n * 8 + 8}, pin_opts);
         QK_CHECK_HIP(hipMemcpyAsync(idx_pin_.data_ptr(),
                                     indices.data_ptr<int64_t>(), n * 8,
                                     hipMemcpyDeviceToHost, stream));
+        int64_t* pinned_n =
+            (int64_t*)((char*)idx_pin_.data_ptr() + idx_pin_.numel() - 8);
+        if (n_dev)
+            QK_CHECK_HIP(hipMemcpyAsync(pinned_n, n_dev, 8,
+                                        hipMemcpyDeviceToHost, stream));
         // waits for the D2H AND the already-launched device-tier kernel
         QK_CHECK_HIP(hipStreamSynchronize(stream));
+        if (n_dev) n = std::min(n, *pinned_n);
         const int64_t* idx = (const int64_t*)idx_pin_.data_ptr();
 
         // parallel count -> chunk prefix -> parallel fill+copy
