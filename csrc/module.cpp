@@ -811,11 +811,14 @@ class ShardTensor {
         }();
         if (mode == 0) return false;
         if (mode == 1) return true;
-        // auto: large training gathers.  The zero-copy kernel's uncached
-        // PCIe reads slow CONCURRENT kernels 5-10x (fabric poisoning,
-        // profiles/timeline_analysis.md), so overlap never pays; the
-        // staged path keeps host traffic on the SDMA engines.
-        return n >= 16384;
+        // default OFF.  Measured trade (profiles/timeline_analysis.md):
+        // the zero-copy kernel's uncached PCIe reads slow concurrent
+        // kernels 5-10x, but the staged path's CPU block + extra hops
+        // cost more end-to-end (products step 5.33 ms staged vs 4.45
+        // zero-copy; feature microbench 38 vs 71 GB/s even with the
+        // persistent pool).  Opt in with QUIVER_STAGED_GATHER=1.
+        (void)n;
+        return false;
     }
 
     // CPU-staged gather of the pinned-host tier (see gather_on).
