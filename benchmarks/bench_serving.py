@@ -241,7 +241,7 @@ def main():
         "metric": "serving-p99-latency",
         "p99_ms": worst_p99,
         "avg_latency_s": avg_lat,
-        "throughput_req_s": agg_tp,
+        "throughput_seeds_s": agg_tp,
         "offered_qps": args.qps,
         "mode": args.mode,
         "request_size": args.request_size,
