@@ -36,7 +36,7 @@ def skewed_batches(nodes, batch, iters, seed=3, alpha=2.0):
 
 
 def bench(shape, cache, policy, device_list, batch=80_000, iters=50,
-          warmup=5, rank=0):
+          warmup=5, rank=0, sorted_ids=False):
     cfg = SHAPES[shape]
     g = torch.Generator().manual_seed(0)
     feat = torch.randn(cfg["nodes"], cfg["dim"], generator=g)
@@ -46,6 +46,8 @@ def bench(shape, cache, policy, device_list, batch=80_000, iters=50,
     row_bytes = cfg["dim"] * 4
     batches = [b.cuda() for b in skewed_batches(cfg["nodes"], batch,
                                                 warmup + iters)]
+    if sorted_ids:
+        batches = [torch.sort(b)[0] for b in batches]
     for i in range(warmup):
         feature[batches[i]]
     torch.cuda.synchronize()
@@ -58,7 +60,7 @@ def bench(shape, cache, policy, device_list, batch=80_000, iters=50,
     dt = time.perf_counter() - t0
     return dict(shape=shape, cache=cache, policy=policy,
                 gbps=total / dt / 1e9, batch=batch, iters=iters,
-                row_bytes=row_bytes)
+                row_bytes=row_bytes, sorted_ids=sorted_ids)
 
 
 if __name__ == "__main__":
@@ -69,6 +71,7 @@ if __name__ == "__main__":
     p.add_argument("--devices", default="0")
     p.add_argument("--batch", type=int, default=80_000)
     p.add_argument("--iters", type=int, default=50)
+    p.add_argument("--sorted", action="store_true")
     args = p.parse_args()
     cfg = SHAPES[args.shape]
     cache = args.cache
@@ -79,5 +82,5 @@ if __name__ == "__main__":
     if args.policy == "p2p_clique_replicate":
         quiver.init_p2p(devices)
     res = bench(args.shape, cache, args.policy, devices, batch=args.batch,
-                iters=args.iters)
+                iters=args.iters, sorted_ids=args.sorted)
     print(json.dumps(res), flush=True)
