@@ -75,6 +75,8 @@ class WorkerPool {
             if (nchunk == 1) fn((int64_t)0);
             return;
         }
+        // one job at a time: concurrent callers would interleave epochs
+        std::lock_guard<std::mutex> serial(run_mu_);
         {
             std::unique_lock<std::mutex> lk(mu_);
             job_ = [&fn](int64_t c) { fn(c); };
@@ -122,6 +124,7 @@ class WorkerPool {
         }
     }
     std::vector<std::thread> workers_;
+    std::mutex run_mu_;
     std::mutex mu_;
     std::condition_variable cv_, done_cv_;
     std::function<void(int64_t)> job_;
