@@ -31,7 +31,7 @@ import torch.distributed as dist
 import torch.nn.functional as F
 
 import quiver
-from quiver.nn import GraphSAGE
+from quiver.nn import GAT, GraphSAGE
 
 # graph presets: ogbn-products (default; the reference's headline E2E
 # config) and ogbn-papers100M (reference benchmarks/ogbn-papers100M/
@@ -95,6 +95,7 @@ def main():
     p.add_argument("--no-overlap", action="store_true",
                    help="disable the sample+gather / compute prefetch overlap")
     p.add_argument("--prefetch-streams", type=int, default=1)
+    p.add_argument("--model", default="sage", choices=["sage", "gat"])
     args = p.parse_args()
 
     ps = PRESETS[args.preset]
@@ -159,8 +160,12 @@ def main():
                              csr_topo=csr_topo)
     feature.from_cpu_tensor(feat_cpu)
 
-    model = GraphSAGE(feat_dim, HIDDEN, n_classes, num_layers=len(FANOUT),
-                      dropout=0.0).to(device)
+    if args.model == "gat":
+        model = GAT(feat_dim, HIDDEN // 4, n_classes,
+                    num_layers=len(FANOUT), heads=4, dropout=0.0).to(device)
+    else:
+        model = GraphSAGE(feat_dim, HIDDEN, n_classes,
+                          num_layers=len(FANOUT), dropout=0.0).to(device)
     if distributed:
         model = torch.nn.parallel.DistributedDataParallel(
             model, device_ids=[local_rank])
@@ -253,7 +258,9 @@ def main():
                     f"{feat_dim} feats, power-law degrees), random-init "
                     "3-layer SAGE",
             "config": {
-                "model": f"graphsage-3L-h{HIDDEN} ogbn-{args.preset}",
+                "model": (f"graphsage-3L-h{HIDDEN}" if args.model == "sage"
+                          else f"gat-3L-h{HIDDEN // 4}x4h")
+                         + f" ogbn-{args.preset}",
                 "global_batch": args.batch * world,
                 "fanout": FANOUT,
                 "parallelism": f"dp{world}",
