@@ -59,3 +59,28 @@ def test_reindex_by_config():
         assert torch.equal(new_feat[order[v]], feat[v])
     # node 0 has the highest degree -> first row after degree sort
     assert order[0].item() == 0
+
+
+def test_topo_clique_discovery(monkeypatch):
+    """Clique grouping from a mocked access matrix: a two-island topology
+    splits, a full mesh stays one clique (no hardcoded splits)."""
+    import quiver.utils as U
+
+    def mesh(n):
+        return [[True] * n for _ in range(n)]
+
+    # full 8-GPU xGMI mesh -> one clique {0..7}
+    monkeypatch.setattr(U, "_access_matrix", lambda devs: mesh(len(devs)))
+    monkeypatch.setattr(U.torch.cuda, "is_available", lambda: True)
+    t = U.Topo(list(range(8)))
+    assert t.p2p_clique_count == 1
+    assert t.p2p_clique(3) == list(range(8))
+
+    # two islands {0,1} and {2,3}
+    island = [[i // 2 == j // 2 for j in range(4)] for i in range(4)]
+    monkeypatch.setattr(U, "_access_matrix", lambda devs: island)
+    t2 = U.Topo([0, 1, 2, 3])
+    assert t2.p2p_clique_count == 2
+    assert t2.p2p_clique(0) == [0, 1]
+    assert t2.p2p_clique(2) == [2, 3]
+    assert t2.get_clique_id(1) == 0 and t2.get_clique_id(3) == 1
