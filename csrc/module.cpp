@@ -1086,6 +1086,59 @@ std::tuple<torch::Tensor, torch::Tensor> wgrad(torch::Tensor a,
     return {c, bias};
 }
 
+// Weighted segment sum (GAT): out[d,h*C+c] = sum_e w[e,h]*x[src[e],h*C+c]
+torch::Tensor segment_wsum(torch::Tensor x, torch::Tensor w,
+                           torch::Tensor src, torch::Tensor dst_ptr,
+                           int64_t heads) {
+    TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kFloat32 && x.dim() == 2);
+    TORCH_CHECK(w.is_cuda() && w.dtype() == torch::kFloat32 && w.dim() == 2 &&
+                w.size(1) == heads);
+    x = x.contiguous();
+    w = w.contiguous();
+    src = src.contiguous();
+    dst_ptr = dst_ptr.contiguous();
+    int64_t n_dst = dst_ptr.numel() - 1;
+    int chead = (int)(x.size(1) / heads);
+    TORCH_CHECK((int64_t)chead * heads == x.size(1), "dim % heads != 0");
+    auto out = torch::empty({n_dst, x.size(1)}, x.options());
+    qk::launch_segment_wsum_fwd(current_stream(), x.data_ptr<float>(),
+                                w.data_ptr<float>(), src.data_ptr<int64_t>(),
+                                dst_ptr.data_ptr<int64_t>(), n_dst,
+                                (int)heads, chead, out.data_ptr<float>());
+    return out;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> segment_wsum_backward(
+    torch::Tensor grad_out, torch::Tensor x, torch::Tensor w,
+    torch::Tensor src, torch::Tensor dst_ptr, int64_t heads,
+    bool need_gx, bool need_gw) {
+    grad_out = grad_out.contiguous();
+    x = x.contiguous();
+    w = w.contiguous();
+    src = src.contiguous();
+    dst_ptr = dst_ptr.contiguous();
+    int64_t n_dst = dst_ptr.numel() - 1;
+    int chead = (int)(x.size(1) / heads);
+    torch::Tensor gx, gw;
+    if (need_gx) {
+        gx = torch::zeros_like(x);
+        qk::launch_segment_wsum_bwd_x(
+            current_stream(), grad_out.data_ptr<float>(),
+            w.data_ptr<float>(), src.data_ptr<int64_t>(),
+            dst_ptr.data_ptr<int64_t>(), n_dst, (int)heads, chead,
+            gx.data_ptr<float>());
+    }
+    if (need_gw) {
+        gw = torch::empty_like(w);
+        qk::launch_segment_wsum_bwd_w(
+            current_stream(), grad_out.data_ptr<float>(),
+            x.data_ptr<float>(), src.data_ptr<int64_t>(),
+            dst_ptr.data_ptr<int64_t>(), n_dst, (int)heads, chead,
+            gw.data_ptr<float>());
+    }
+    return {gx, gw};
+}
+
 void init_p2p(const std::vector<int>& devices) {
     // On an 8x MI355X node every pair is xGMI-connected: enable the full
     // clique (reference init_p2p, quiver_feature.cu:378-421; no NVLink-style
@@ -1270,6 +1323,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("segment_mean_gather_backward", &segment_mean_gather_backward,
           py::call_guard<py::gil_scoped_release>());
 
+    m.def("segment_wsum", &segment_wsum,
+          "weighted segment sum over dst-sorted edges (GAT aggregation)");
+    m.def("segment_wsum_backward", &segment_wsum_backward);
     m.def("wgrad", &wgrad,
           "tall-skinny A^T@B (+ optional A column sums) via split-K atomics");
     m.def("create_nccl_id", &create_nccl_id);

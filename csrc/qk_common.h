@@ -152,4 +152,21 @@ void launch_segment_mean_bwd(hipStream_t s, const float* grad_out,
                              const int64_t* src, const int64_t* dst_ptr,
                              int64_t n_dst, int64_t dim, float* grad_x);
 
+// Weighted segment sum (GAT attention aggregation), x [n_src, H*C],
+// w [n_edges, H] (per-edge per-head attention):
+//   out[d, h*C+c] = sum_{e in segment(d)} w[e,h] * x[src[e], h*C+c]
+void launch_segment_wsum_fwd(hipStream_t s, const float* x, const float* w,
+                             const int64_t* src, const int64_t* dst_ptr,
+                             int64_t n_dst, int heads, int chead, float* out);
+// grad_x[src[e], h*C+c] += w[e,h] * grad_out[d, h*C+c]  (pre-zeroed)
+void launch_segment_wsum_bwd_x(hipStream_t s, const float* grad_out,
+                               const float* w, const int64_t* src,
+                               const int64_t* dst_ptr, int64_t n_dst,
+                               int heads, int chead, float* grad_x);
+// grad_w[e,h] = sum_c grad_out[d, h*C+c] * x[src[e], h*C+c]
+void launch_segment_wsum_bwd_w(hipStream_t s, const float* grad_out,
+                               const float* x, const int64_t* src,
+                               const int64_t* dst_ptr, int64_t n_dst,
+                               int heads, int chead, float* grad_w);
+
 }  // namespace qk

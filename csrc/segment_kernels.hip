@@ -80,6 +80,104 @@ segment_mean_bwd_kernel(const float* __restrict__ grad_out,
     }
 }
 
+// ---- weighted segment sum (GAT attention aggregation) -------------------
+// dim = heads*chead; weights indexed per (edge, head).  Lanes stride the
+// feature dim; head of a channel c is c / chead (chead is a multiple of
+// VPL in the models here, so a VPL chunk never straddles heads — guarded
+// by the launcher).
+
+__global__ void __launch_bounds__(BLOCK)
+segment_wsum_fwd_kernel(const float* __restrict__ x,
+                        const float* __restrict__ w,
+                        const int64_t* __restrict__ src,
+                        const int64_t* __restrict__ dst_ptr, int64_t n_dst,
+                        int heads, int chead, float* __restrict__ out) {
+    const int64_t dim = (int64_t)heads * chead;
+    const int sub_id = threadIdx.x / SUB;
+    const int lane = threadIdx.x % SUB;
+    int64_t d = (int64_t)blockIdx.x * ROWS_PER_BLOCK + sub_id;
+    const int64_t stride = (int64_t)gridDim.x * ROWS_PER_BLOCK;
+    for (; d < n_dst; d += stride) {
+        const int64_t beg = dst_ptr[d], end = dst_ptr[d + 1];
+        for (int64_t c = (int64_t)lane * VPL; c < dim; c += SUB * VPL) {
+            const int h = (int)(c / chead);
+            float acc[VPL] = {0.f, 0.f, 0.f, 0.f};
+            const int wd = (int)min((int64_t)VPL, dim - c);
+            for (int64_t e = beg; e < end; ++e) {
+                const float we = w[e * heads + h];
+                const float* row = x + src[e] * dim + c;
+                if (wd == VPL) {
+                    const float4 v = *reinterpret_cast<const float4*>(row);
+                    acc[0] += we * v.x; acc[1] += we * v.y;
+                    acc[2] += we * v.z; acc[3] += we * v.w;
+                } else {
+                    for (int q = 0; q < wd; ++q) acc[q] += we * row[q];
+                }
+            }
+            float* orow = out + d * dim + c;
+            if (wd == VPL) {
+                float4 v{acc[0], acc[1], acc[2], acc[3]};
+                *reinterpret_cast<float4*>(orow) = v;
+            } else {
+                for (int q = 0; q < wd; ++q) orow[q] = acc[q];
+            }
+        }
+    }
+}
+
+__global__ void __launch_bounds__(BLOCK)
+segment_wsum_bwd_x_kernel(const float* __restrict__ grad_out,
+                          const float* __restrict__ w,
+                          const int64_t* __restrict__ src,
+                          const int64_t* __restrict__ dst_ptr, int64_t n_dst,
+                          int heads, int chead, float* __restrict__ grad_x) {
+    const int64_t dim = (int64_t)heads * chead;
+    const int sub_id = threadIdx.x / SUB;
+    const int lane = threadIdx.x % SUB;
+    int64_t d = (int64_t)blockIdx.x * ROWS_PER_BLOCK + sub_id;
+    const int64_t stride = (int64_t)gridDim.x * ROWS_PER_BLOCK;
+    for (; d < n_dst; d += stride) {
+        const int64_t beg = dst_ptr[d], end = dst_ptr[d + 1];
+        const float* orow = grad_out + d * dim;
+        for (int64_t e = beg; e < end; ++e) {
+            float* grow = grad_x + src[e] * dim;
+            const float* wrow = w + e * heads;
+            for (int64_t c = lane; c < dim; c += SUB)
+                atomicAdd(&grow[c], wrow[c / chead] * orow[c]);
+        }
+    }
+}
+
+__global__ void __launch_bounds__(BLOCK)
+segment_wsum_bwd_w_kernel(const float* __restrict__ grad_out,
+                          const float* __restrict__ x,
+                          const int64_t* __restrict__ src,
+                          const int64_t* __restrict__ dst_ptr, int64_t n_dst,
+                          int heads, int chead, float* __restrict__ grad_w) {
+    // one subgroup per dst; lanes cooperate per edge-head dot over chead
+    const int64_t dim = (int64_t)heads * chead;
+    const int sub_id = threadIdx.x / SUB;
+    const int lane = threadIdx.x % SUB;
+    int64_t d = (int64_t)blockIdx.x * ROWS_PER_BLOCK + sub_id;
+    const int64_t stride = (int64_t)gridDim.x * ROWS_PER_BLOCK;
+    for (; d < n_dst; d += stride) {
+        const int64_t beg = dst_ptr[d], end = dst_ptr[d + 1];
+        const float* orow = grad_out + d * dim;
+        for (int64_t e = beg; e < end; ++e) {
+            const float* xrow = x + src[e] * dim;
+            for (int h = 0; h < heads; ++h) {
+                float acc = 0.f;
+                const int64_t base = (int64_t)h * chead;
+                for (int c = lane; c < chead; c += SUB)
+                    acc += orow[base + c] * xrow[base + c];
+                for (int off = SUB / 2; off > 0; off >>= 1)
+                    acc += __shfl_down(acc, off, SUB);
+                if (lane == 0) grad_w[e * heads + h] = acc;
+            }
+        }
+    }
+}
+
 inline int grid_for(int64_t work, int per_block) {
     int64_t blocks = (work + per_block - 1) / per_block;
     if (blocks > 2048) blocks = 2048;
@@ -104,6 +202,45 @@ void launch_segment_mean_bwd(hipStream_t s, const float* grad_out,
     if (n_dst == 0) return;
     segment_mean_bwd_kernel<<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0, s>>>(
         grad_out, src, dst_ptr, n_dst, dim, grad_x);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+static void check_chead(int chead) {
+    if (chead % VPL != 0)
+        throw std::runtime_error(
+            "segment_wsum: per-head channels must be a multiple of 4 "
+            "so vector chunks stay within one head");
+}
+
+void launch_segment_wsum_fwd(hipStream_t s, const float* x, const float* w,
+                             const int64_t* src, const int64_t* dst_ptr,
+                             int64_t n_dst, int heads, int chead, float* out) {
+    if (n_dst == 0) return;
+    check_chead(chead);
+    segment_wsum_fwd_kernel<<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0, s>>>(
+        x, w, src, dst_ptr, n_dst, heads, chead, out);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+void launch_segment_wsum_bwd_x(hipStream_t s, const float* grad_out,
+                               const float* w, const int64_t* src,
+                               const int64_t* dst_ptr, int64_t n_dst,
+                               int heads, int chead, float* grad_x) {
+    if (n_dst == 0) return;
+    segment_wsum_bwd_x_kernel<<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0,
+                                s>>>(grad_out, w, src, dst_ptr, n_dst, heads,
+                                     chead, grad_x);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+void launch_segment_wsum_bwd_w(hipStream_t s, const float* grad_out,
+                               const float* x, const int64_t* src,
+                               const int64_t* dst_ptr, int64_t n_dst,
+                               int heads, int chead, float* grad_w) {
+    if (n_dst == 0) return;
+    segment_wsum_bwd_w_kernel<<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0,
+                                s>>>(grad_out, x, src, dst_ptr, n_dst, heads,
+                                     chead, grad_w);
     QK_CHECK_HIP(hipGetLastError());
 }
 

@@ -47,6 +47,28 @@ class _SegmentMeanAgg(torch.autograd.Function):
         return grad_x, None, None
 
 
+class _SegmentWSum(torch.autograd.Function):
+    """Weighted segment sum over dst-sorted edges (GAT attention
+    aggregation, csrc/segment_kernels.hip).  Replaces the torch chain
+    h_src[src] (materialize [E,H,C]) * alpha -> zeros -> index_add, which
+    moves ~E*H*C*4 bytes twice per layer."""
+
+    @staticmethod
+    def forward(ctx, x, w, src, dst_ptr, heads):
+        ctx.save_for_backward(x, w, src, dst_ptr)
+        ctx.heads = heads
+        return _ext.segment_wsum(x, w, src, dst_ptr, heads)
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        x, w, src, dst_ptr = ctx.saved_tensors
+        gx, gw = _ext.segment_wsum_backward(
+            grad_out.contiguous(), x, w, src, dst_ptr, ctx.heads,
+            ctx.needs_input_grad[0], ctx.needs_input_grad[1])
+        return (gx if ctx.needs_input_grad[0] else None,
+                gw if ctx.needs_input_grad[1] else None, None, None, None)
+
+
 class _QLinearFn(torch.autograd.Function):
     """Linear layer whose weight gradient uses the split-K HIP kernel
     (csrc/wgrad_kernels.hip).
@@ -162,8 +184,10 @@ class GATConv(nn.Module):
     """Graph attention convolution (multi-head, concat)."""
 
     def __init__(self, in_channels, out_channels, heads=1, concat=True,
-                 negative_slope=0.2, dropout=0.0, bias=True):
+                 negative_slope=0.2, dropout=0.0, bias=True,
+                 sorted_dst=False):
         super().__init__()
+        self.sorted_dst = sorted_dst
         self.heads = heads
         self.out_channels = out_channels
         self.concat = concat
@@ -208,10 +232,21 @@ class GATConv(nn.Module):
         if self.training and self.dropout > 0:
             alpha = F.dropout(alpha, p=self.dropout)
 
-        msg = h_src[src] * alpha.unsqueeze(-1)      # [E, H, C]
-        out = torch.zeros((n_dst, H, C), dtype=msg.dtype, device=msg.device)
-        out.index_add_(0, dst, msg)
-        out = out.reshape(n_dst, H * C) if self.concat else out.mean(1)
+        if (self.sorted_dst and h_src.is_cuda
+                and h_src.dtype == torch.float32 and C % 4 == 0
+                and dst.numel() > 0):
+            # fused weighted segment sum over the dst-sorted edges
+            dst_ptr = torch.searchsorted(dst, _arange(n_dst + 1, dst.device))
+            out = _SegmentWSum.apply(h_src.reshape(-1, H * C), alpha, src,
+                                     dst_ptr, H)
+            out = out if self.concat else out.view(n_dst, H, C).mean(1)
+        else:
+            msg = h_src[src] * alpha.unsqueeze(-1)      # [E, H, C]
+            out = torch.zeros((n_dst, H, C), dtype=msg.dtype,
+                              device=msg.device)
+            out.index_add_(0, dst, msg)
+            out = (out.reshape(n_dst, H * C) if self.concat
+                   else out.mean(1))
         if self.bias is not None:
             out = out + self.bias
         return out
@@ -264,21 +299,24 @@ class GraphSAGE(nn.Module):
 
 class GAT(nn.Module):
     def __init__(self, in_channels, hidden_channels, out_channels,
-                 num_layers=2, heads=4, dropout=0.5):
+                 num_layers=2, heads=4, dropout=0.5, sorted_dst=True):
         super().__init__()
         self.num_layers = num_layers
         self.dropout = dropout
         self.convs = nn.ModuleList()
+        # sorted_dst: GraphSageSampler emits dst-sorted adjs -> fused kernel
         if num_layers == 1:
-            self.convs.append(GATConv(in_channels, out_channels, heads=1))
+            self.convs.append(GATConv(in_channels, out_channels, heads=1,
+                                      sorted_dst=sorted_dst))
         else:
             self.convs.append(GATConv(in_channels, hidden_channels,
-                                      heads=heads))
+                                      heads=heads, sorted_dst=sorted_dst))
             for _ in range(num_layers - 2):
                 self.convs.append(GATConv(hidden_channels * heads,
-                                          hidden_channels, heads=heads))
+                                          hidden_channels, heads=heads,
+                                          sorted_dst=sorted_dst))
             self.convs.append(GATConv(hidden_channels * heads, out_channels,
-                                      heads=1))
+                                      heads=1, sorted_dst=sorted_dst))
 
     def forward(self, x, adjs):
         for i, (edge_index, _, size) in enumerate(adjs):
