@@ -193,7 +193,7 @@ class GATConv(nn.Module):
         self.concat = concat
         self.negative_slope = negative_slope
         self.dropout = dropout
-        self.lin = nn.Linear(in_channels, heads * out_channels, bias=False)
+        self.lin = QLinear(in_channels, heads * out_channels, bias=False)
         self.att_src = nn.Parameter(torch.empty(1, heads, out_channels))
         self.att_dst = nn.Parameter(torch.empty(1, heads, out_channels))
         out_dim = heads * out_channels if concat else out_channels
@@ -214,7 +214,13 @@ class GATConv(nn.Module):
         n_dst = x_dst.size(0) if size is None else int(size[1])
 
         h_src = self.lin(x_src).view(-1, H, C)
-        h_dst = self.lin(x_dst).view(-1, H, C)
+        if (x_dst.data_ptr() == x_src.data_ptr()
+                and x_dst.size(0) <= x_src.size(0)):
+            # bipartite prefix convention: x_dst is x_src[:n_dst] — reuse
+            # the projection instead of a second GEMM (+ its backward)
+            h_dst = h_src[:x_dst.size(0)]
+        else:
+            h_dst = self.lin(x_dst).view(-1, H, C)
         alpha_src = (h_src * self.att_src).sum(-1)  # [N_src, H]
         alpha_dst = (h_dst * self.att_dst).sum(-1)  # [N_dst, H]
         alpha = alpha_src[src] + alpha_dst[dst]     # [E, H]
