@@ -1139,6 +1139,37 @@ std::tuple<torch::Tensor, torch::Tensor> segment_wsum_backward(
     return {gx, gw};
 }
 
+// Numerically-stable segment softmax over [E, H] (dst_ptr segments).
+torch::Tensor segment_softmax(torch::Tensor a, torch::Tensor dst_ptr,
+                              int64_t heads) {
+    TORCH_CHECK(a.is_cuda() && a.dtype() == torch::kFloat32 && a.dim() == 2 &&
+                a.size(1) == heads);
+    a = a.contiguous();
+    dst_ptr = dst_ptr.contiguous();
+    int64_t n_dst = dst_ptr.numel() - 1;
+    auto out = torch::empty_like(a);
+    qk::launch_segment_softmax_fwd(current_stream(), a.data_ptr<float>(),
+                                   dst_ptr.data_ptr<int64_t>(), n_dst,
+                                   (int)heads, out.data_ptr<float>());
+    return out;
+}
+
+torch::Tensor segment_softmax_backward(torch::Tensor grad_out,
+                                       torch::Tensor out,
+                                       torch::Tensor dst_ptr, int64_t heads) {
+    grad_out = grad_out.contiguous();
+    out = out.contiguous();
+    dst_ptr = dst_ptr.contiguous();
+    int64_t n_dst = dst_ptr.numel() - 1;
+    auto ga = torch::empty_like(out);
+    qk::launch_segment_softmax_bwd(current_stream(),
+                                   grad_out.data_ptr<float>(),
+                                   out.data_ptr<float>(),
+                                   dst_ptr.data_ptr<int64_t>(), n_dst,
+                                   (int)heads, ga.data_ptr<float>());
+    return ga;
+}
+
 void init_p2p(const std::vector<int>& devices) {
     // On an 8x MI355X node every pair is xGMI-connected: enable the full
     // clique (reference init_p2p, quiver_feature.cu:378-421; no NVLink-style
@@ -1323,6 +1354,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("segment_mean_gather_backward", &segment_mean_gather_backward,
           py::call_guard<py::gil_scoped_release>());
 
+    m.def("segment_softmax", &segment_softmax,
+          "numerically-stable softmax over dst-sorted edge segments");
+    m.def("segment_softmax_backward", &segment_softmax_backward);
     m.def("segment_wsum", &segment_wsum,
           "weighted segment sum over dst-sorted edges (GAT aggregation)");
     m.def("segment_wsum_backward", &segment_wsum_backward);

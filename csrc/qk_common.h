@@ -169,4 +169,15 @@ void launch_segment_wsum_bwd_w(hipStream_t s, const float* grad_out,
                                const int64_t* dst_ptr, int64_t n_dst,
                                int heads, int chead, float* grad_w);
 
+// Segment softmax over [n_edges, H] grouped by dst_ptr segments:
+//   out[e,h] = exp(a[e,h] - max_seg) / sum_seg exp(...)
+// (numerically-stable; empty segments produce nothing)
+void launch_segment_softmax_fwd(hipStream_t s, const float* a,
+                                const int64_t* dst_ptr, int64_t n_dst,
+                                int heads, float* out);
+// g_a[e,h] = out[e,h] * (g[e,h] - sum_seg g*out)
+void launch_segment_softmax_bwd(hipStream_t s, const float* grad_out,
+                                const float* out, const int64_t* dst_ptr,
+                                int64_t n_dst, int heads, float* grad_a);
+
 }  // namespace qk

@@ -178,6 +178,57 @@ segment_wsum_bwd_w_kernel(const float* __restrict__ grad_out,
     }
 }
 
+// ---- segment softmax ([E,H] grouped by dst segments) --------------------
+// One thread per (dst, head): segments are fanout-sized (<= ~25), so a
+// scalar two-pass (max, exp-sum) loop is cheap and avoids the sort-based
+// torch scatter_reduce lowering (~120 rocprim kernels per step).
+
+__global__ void __launch_bounds__(BLOCK)
+segment_softmax_fwd_kernel(const float* __restrict__ a,
+                           const int64_t* __restrict__ dst_ptr,
+                           int64_t n_dst, int heads, float* __restrict__ out) {
+    int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    const int64_t total = n_dst * heads;
+    for (; t < total; t += stride) {
+        const int64_t d = t / heads;
+        const int h = (int)(t % heads);
+        const int64_t beg = dst_ptr[d], end = dst_ptr[d + 1];
+        if (end <= beg) continue;
+        float m = -INFINITY;
+        for (int64_t e = beg; e < end; ++e)
+            m = fmaxf(m, a[e * heads + h]);
+        float s = 0.f;
+        for (int64_t e = beg; e < end; ++e)
+            s += __expf(a[e * heads + h] - m);
+        const float inv = 1.0f / fmaxf(s, 1e-16f);
+        for (int64_t e = beg; e < end; ++e)
+            out[e * heads + h] = __expf(a[e * heads + h] - m) * inv;
+    }
+}
+
+__global__ void __launch_bounds__(BLOCK)
+segment_softmax_bwd_kernel(const float* __restrict__ grad_out,
+                           const float* __restrict__ out,
+                           const int64_t* __restrict__ dst_ptr,
+                           int64_t n_dst, int heads,
+                           float* __restrict__ grad_a) {
+    int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    const int64_t total = n_dst * heads;
+    for (; t < total; t += stride) {
+        const int64_t d = t / heads;
+        const int h = (int)(t % heads);
+        const int64_t beg = dst_ptr[d], end = dst_ptr[d + 1];
+        float s = 0.f;
+        for (int64_t e = beg; e < end; ++e)
+            s += grad_out[e * heads + h] * out[e * heads + h];
+        for (int64_t e = beg; e < end; ++e)
+            grad_a[e * heads + h] =
+                out[e * heads + h] * (grad_out[e * heads + h] - s);
+    }
+}
+
 inline int grid_for(int64_t work, int per_block) {
     int64_t blocks = (work + per_block - 1) / per_block;
     if (blocks > 2048) blocks = 2048;
@@ -241,6 +292,25 @@ void launch_segment_wsum_bwd_w(hipStream_t s, const float* grad_out,
     segment_wsum_bwd_w_kernel<<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0,
                                 s>>>(grad_out, x, src, dst_ptr, n_dst, heads,
                                      chead, grad_w);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+void launch_segment_softmax_fwd(hipStream_t s, const float* a,
+                                const int64_t* dst_ptr, int64_t n_dst,
+                                int heads, float* out) {
+    if (n_dst == 0) return;
+    segment_softmax_fwd_kernel<<<grid_for(n_dst * heads, BLOCK), BLOCK, 0,
+                                 s>>>(a, dst_ptr, n_dst, heads, out);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+void launch_segment_softmax_bwd(hipStream_t s, const float* grad_out,
+                                const float* out, const int64_t* dst_ptr,
+                                int64_t n_dst, int heads, float* grad_a) {
+    if (n_dst == 0) return;
+    segment_softmax_bwd_kernel<<<grid_for(n_dst * heads, BLOCK), BLOCK, 0,
+                                 s>>>(grad_out, out, dst_ptr, n_dst, heads,
+                                      grad_a);
     QK_CHECK_HIP(hipGetLastError());
 }
 

@@ -69,6 +69,26 @@ class _SegmentWSum(torch.autograd.Function):
                 gw if ctx.needs_input_grad[1] else None, None, None, None)
 
 
+class _SegmentSoftmax(torch.autograd.Function):
+    """Softmax over dst-sorted edge segments (csrc/segment_kernels.hip).
+    torch's scatter_reduce(amax) path lowers to ~120 rocprim sort kernels
+    per step on ROCm; this is two scalar kernels."""
+
+    @staticmethod
+    def forward(ctx, a, dst_ptr, heads):
+        out = _ext.segment_softmax(a, dst_ptr, heads)
+        ctx.save_for_backward(out, dst_ptr)
+        ctx.heads = heads
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        out, dst_ptr = ctx.saved_tensors
+        return (_ext.segment_softmax_backward(grad_out.contiguous(), out,
+                                              dst_ptr, ctx.heads),
+                None, None)
+
+
 class _QLinearFn(torch.autograd.Function):
     """Linear layer whose weight gradient uses the split-K HIP kernel
     (csrc/wgrad_kernels.hip).
@@ -225,16 +245,23 @@ class GATConv(nn.Module):
         alpha_dst = (h_dst * self.att_dst).sum(-1)  # [N_dst, H]
         alpha = alpha_src[src] + alpha_dst[dst]     # [E, H]
         alpha = F.leaky_relu(alpha, self.negative_slope)
-        # segment softmax over incoming edges of each dst
-        alpha_max = torch.full((n_dst, H), float("-inf"), dtype=alpha.dtype,
-                               device=alpha.device)
-        alpha_max = alpha_max.scatter_reduce(0, dst.unsqueeze(-1).expand_as(
-            alpha), alpha, reduce="amax", include_self=True)
-        alpha = (alpha - alpha_max[dst]).exp()
-        denom = torch.zeros((n_dst, H), dtype=alpha.dtype,
-                            device=alpha.device)
-        denom.index_add_(0, dst, alpha)
-        alpha = alpha / denom[dst].clamp(min=1e-16)
+        fused = (self.sorted_dst and alpha.is_cuda
+                 and alpha.dtype == torch.float32 and dst.numel() > 0)
+        if fused:
+            dst_ptr = torch.searchsorted(dst, _arange(n_dst + 1, dst.device))
+            alpha = _SegmentSoftmax.apply(alpha, dst_ptr, H)
+        else:
+            # segment softmax over incoming edges of each dst (torch path)
+            alpha_max = torch.full((n_dst, H), float("-inf"),
+                                   dtype=alpha.dtype, device=alpha.device)
+            alpha_max = alpha_max.scatter_reduce(
+                0, dst.unsqueeze(-1).expand_as(alpha), alpha, reduce="amax",
+                include_self=True)
+            alpha = (alpha - alpha_max[dst]).exp()
+            denom = torch.zeros((n_dst, H), dtype=alpha.dtype,
+                                device=alpha.device)
+            denom.index_add_(0, dst, alpha)
+            alpha = alpha / denom[dst].clamp(min=1e-16)
         if self.training and self.dropout > 0:
             alpha = F.dropout(alpha, p=self.dropout)
 
@@ -242,7 +269,9 @@ class GATConv(nn.Module):
                 and h_src.dtype == torch.float32 and C % 4 == 0
                 and dst.numel() > 0):
             # fused weighted segment sum over the dst-sorted edges
-            dst_ptr = torch.searchsorted(dst, _arange(n_dst + 1, dst.device))
+            if not fused:
+                dst_ptr = torch.searchsorted(dst,
+                                             _arange(n_dst + 1, dst.device))
             out = _SegmentWSum.apply(h_src.reshape(-1, H * C), alpha, src,
                                      dst_ptr, H)
             out = out if self.concat else out.view(n_dst, H, C).mean(1)

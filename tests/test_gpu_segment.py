@@ -130,3 +130,36 @@ def test_gat_fused_autograd_matches_fallback():
     assert torch.allclose(xf.grad, xr.grad, atol=1e-4)
     for pf, pr in zip(conv_f.parameters(), conv_r.parameters()):
         assert torch.allclose(pf.grad, pr.grad, atol=1e-3)
+
+
+def test_segment_softmax_matches_torch():
+    from quiver import _ext
+    g = torch.Generator().manual_seed(2)
+    n_dst, H = 70, 4
+    deg = torch.randint(0, 9, (n_dst,), generator=g)
+    dst_ptr = torch.zeros(n_dst + 1, dtype=torch.long)
+    dst_ptr[1:] = deg.cumsum(0)
+    E = int(dst_ptr[-1])
+    dst = torch.repeat_interleave(torch.arange(n_dst), deg)
+    a = torch.randn(E, H, generator=g)
+
+    ac = a.cuda().requires_grad_(True)
+    out = _ext.segment_softmax(ac, dst_ptr.cuda(), H)
+
+    # torch reference
+    a2 = a.cuda().requires_grad_(True)
+    amax = torch.full((n_dst, H), float("-inf"), device="cuda")
+    amax = amax.scatter_reduce(0, dst.cuda().unsqueeze(-1).expand(E, H), a2,
+                               reduce="amax", include_self=True)
+    ex = (a2 - amax[dst.cuda()]).exp()
+    den = torch.zeros(n_dst, H, device="cuda")
+    den.index_add_(0, dst.cuda(), ex)
+    want = ex / den[dst.cuda()].clamp(min=1e-16)
+    assert torch.allclose(out, want, atol=1e-5), \
+        (out - want).abs().max().item()
+
+    go = torch.randn(E, H, generator=g).cuda()
+    ga = _ext.segment_softmax_backward(go, out, dst_ptr.cuda(), H)
+    want.backward(go)
+    assert torch.allclose(ga, a2.grad, atol=1e-4), \
+        (ga - a2.grad).abs().max().item()
