@@ -38,12 +38,15 @@ from quiver.nn import GAT, GraphSAGE
 # dist_sampling_ogb_paper100M_quiver.py:128,192-197 — same fanout/batch,
 # 8G cache, 128 feats; host-DRAM feature spill is the point of the config)
 PRESETS = {
-    # products keeps the reference's exact headline config (UVA-sampled
-    # graph); papers100M defaults to the HBM-resident graph — the 12.5 GB
-    # of CSR columns trivially fit in 288 GB HBM3E, which is the
-    # MI355X-idiomatic placement (features still spill to host DRAM).
+    # Both presets default to an HBM-resident graph: the CSR columns
+    # (products 1 GB, papers100M 12.5 GB) trivially fit in 288 GB HBM3E —
+    # the MI355X-idiomatic placement.  The FEATURE store keeps the named
+    # configs' layout (hot cache in HBM, cold tier zero-copy/UVA in host
+    # DRAM).  `--mode UVA` additionally keeps the graph in pinned host
+    # memory (the reference's small-GPU setup); both are measured in
+    # profiles/SUMMARY.md.
     "products": dict(nodes=2_449_029, edges=123_718_280, feat_dim=100,
-                     classes=47, train=196_615, cache="196M", mode="UVA"),
+                     classes=47, train=196_615, cache="196M", mode="GPU"),
     "papers100M": dict(nodes=111_059_956, edges=1_615_685_872, feat_dim=128,
                        classes=172, train=1_207_179, cache="8G", mode="GPU"),
 }
