@@ -1170,6 +1170,50 @@ torch::Tensor segment_softmax_backward(torch::Tensor grad_out,
     return ga;
 }
 
+// Fused GAT attention coefficients (gather + add + leaky_relu + segment
+// softmax in one pass; no [E,H] intermediates materialized by torch).
+torch::Tensor gat_alpha(torch::Tensor asrc, torch::Tensor adst,
+                        torch::Tensor src, torch::Tensor dst_ptr,
+                        int64_t heads, double slope, int64_t n_edges) {
+    TORCH_CHECK(asrc.is_cuda() && asrc.dtype() == torch::kFloat32 &&
+                adst.dtype() == torch::kFloat32 &&
+                asrc.size(1) == heads && adst.size(1) == heads);
+    asrc = asrc.contiguous();
+    adst = adst.contiguous();
+    src = src.contiguous();
+    dst_ptr = dst_ptr.contiguous();
+    int64_t n_dst = dst_ptr.numel() - 1;
+    auto alpha = torch::empty({n_edges, heads}, asrc.options());
+    qk::launch_gat_alpha_fwd(current_stream(), asrc.data_ptr<float>(),
+                             adst.data_ptr<float>(),
+                             src.data_ptr<int64_t>(),
+                             dst_ptr.data_ptr<int64_t>(), n_dst, (int)heads,
+                             (float)slope, alpha.data_ptr<float>());
+    return alpha;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> gat_alpha_backward(
+    torch::Tensor grad_alpha, torch::Tensor alpha, torch::Tensor asrc,
+    torch::Tensor adst, torch::Tensor src, torch::Tensor dst_ptr,
+    int64_t heads, double slope) {
+    grad_alpha = grad_alpha.contiguous();
+    alpha = alpha.contiguous();
+    asrc = asrc.contiguous();
+    adst = adst.contiguous();
+    src = src.contiguous();
+    dst_ptr = dst_ptr.contiguous();
+    int64_t n_dst = dst_ptr.numel() - 1;
+    auto g_asrc = torch::zeros_like(asrc);
+    auto g_adst = torch::zeros_like(adst);
+    qk::launch_gat_alpha_bwd(
+        current_stream(), grad_alpha.data_ptr<float>(),
+        alpha.data_ptr<float>(), asrc.data_ptr<float>(),
+        adst.data_ptr<float>(), src.data_ptr<int64_t>(),
+        dst_ptr.data_ptr<int64_t>(), n_dst, (int)heads, (float)slope,
+        g_asrc.data_ptr<float>(), g_adst.data_ptr<float>());
+    return {g_asrc, g_adst};
+}
+
 void init_p2p(const std::vector<int>& devices) {
     // On an 8x MI355X node every pair is xGMI-connected: enable the full
     // clique (reference init_p2p, quiver_feature.cu:378-421; no NVLink-style
@@ -1354,6 +1398,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("segment_mean_gather_backward", &segment_mean_gather_backward,
           py::call_guard<py::gil_scoped_release>());
 
+    m.def("gat_alpha", &gat_alpha,
+          "fused GAT attention coefficients over dst-sorted edges");
+    m.def("gat_alpha_backward", &gat_alpha_backward);
     m.def("segment_softmax", &segment_softmax,
           "numerically-stable softmax over dst-sorted edge segments");
     m.def("segment_softmax_backward", &segment_softmax_backward);

@@ -180,4 +180,19 @@ void launch_segment_softmax_bwd(hipStream_t s, const float* grad_out,
                                 const float* out, const int64_t* dst_ptr,
                                 int64_t n_dst, int heads, float* grad_a);
 
+// Fully fused GAT attention coefficients: from per-node logits to
+// normalized per-edge attention in one kernel,
+//   alpha[e,h] = softmax_seg(leaky_relu(asrc[src[e],h] + adst[d,h]))
+void launch_gat_alpha_fwd(hipStream_t s, const float* asrc,
+                          const float* adst, const int64_t* src,
+                          const int64_t* dst_ptr, int64_t n_dst, int heads,
+                          float slope, float* alpha);
+// backward through softmax + leaky_relu + the two logit gathers:
+// g_asrc accumulated with atomics (pre-zeroed), g_adst written directly.
+void launch_gat_alpha_bwd(hipStream_t s, const float* grad_alpha,
+                          const float* alpha, const float* asrc,
+                          const float* adst, const int64_t* src,
+                          const int64_t* dst_ptr, int64_t n_dst, int heads,
+                          float slope, float* g_asrc, float* g_adst);
+
 }  // namespace qk

@@ -229,6 +229,76 @@ segment_softmax_bwd_kernel(const float* __restrict__ grad_out,
     }
 }
 
+// ---- fully fused GAT attention coefficients -----------------------------
+// One thread per (dst, head).  Segments are fanout-sized; the
+// pre-activation is recomputed per pass (asrc/adst stay L2-hot) instead
+// of materializing any [E,H] intermediate.
+
+__device__ __forceinline__ float lrelu(float v, float slope) {
+    return v > 0.f ? v : v * slope;
+}
+
+__global__ void __launch_bounds__(BLOCK)
+gat_alpha_fwd_kernel(const float* __restrict__ asrc,
+                     const float* __restrict__ adst,
+                     const int64_t* __restrict__ src,
+                     const int64_t* __restrict__ dst_ptr, int64_t n_dst,
+                     int heads, float slope, float* __restrict__ alpha) {
+    int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    const int64_t total = n_dst * heads;
+    for (; t < total; t += stride) {
+        const int64_t d = t / heads;
+        const int h = (int)(t % heads);
+        const int64_t beg = dst_ptr[d], end = dst_ptr[d + 1];
+        if (end <= beg) continue;
+        const float ad = adst[d * heads + h];
+        float m = -INFINITY;
+        for (int64_t e = beg; e < end; ++e)
+            m = fmaxf(m, lrelu(asrc[src[e] * heads + h] + ad, slope));
+        float s = 0.f;
+        for (int64_t e = beg; e < end; ++e)
+            s += __expf(lrelu(asrc[src[e] * heads + h] + ad, slope) - m);
+        const float inv = 1.0f / fmaxf(s, 1e-16f);
+        for (int64_t e = beg; e < end; ++e)
+            alpha[e * heads + h] =
+                __expf(lrelu(asrc[src[e] * heads + h] + ad, slope) - m) * inv;
+    }
+}
+
+__global__ void __launch_bounds__(BLOCK)
+gat_alpha_bwd_kernel(const float* __restrict__ grad_alpha,
+                     const float* __restrict__ alpha,
+                     const float* __restrict__ asrc,
+                     const float* __restrict__ adst,
+                     const int64_t* __restrict__ src,
+                     const int64_t* __restrict__ dst_ptr, int64_t n_dst,
+                     int heads, float slope, float* __restrict__ g_asrc,
+                     float* __restrict__ g_adst) {
+    int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    const int64_t total = n_dst * heads;
+    for (; t < total; t += stride) {
+        const int64_t d = t / heads;
+        const int h = (int)(t % heads);
+        const int64_t beg = dst_ptr[d], end = dst_ptr[d + 1];
+        const float ad = adst[d * heads + h];
+        float S = 0.f;
+        for (int64_t e = beg; e < end; ++e)
+            S += grad_alpha[e * heads + h] * alpha[e * heads + h];
+        float acc = 0.f;
+        for (int64_t e = beg; e < end; ++e) {
+            const float gsm = alpha[e * heads + h] *
+                              (grad_alpha[e * heads + h] - S);
+            const float pre = asrc[src[e] * heads + h] + ad;
+            const float gpre = pre > 0.f ? gsm : gsm * slope;
+            atomicAdd(&g_asrc[src[e] * heads + h], gpre);
+            acc += gpre;
+        }
+        g_adst[d * heads + h] = acc;
+    }
+}
+
 inline int grid_for(int64_t work, int per_block) {
     int64_t blocks = (work + per_block - 1) / per_block;
     if (blocks > 2048) blocks = 2048;
@@ -311,6 +381,28 @@ void launch_segment_softmax_bwd(hipStream_t s, const float* grad_out,
     segment_softmax_bwd_kernel<<<grid_for(n_dst * heads, BLOCK), BLOCK, 0,
                                  s>>>(grad_out, out, dst_ptr, n_dst, heads,
                                       grad_a);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+void launch_gat_alpha_fwd(hipStream_t s, const float* asrc,
+                          const float* adst, const int64_t* src,
+                          const int64_t* dst_ptr, int64_t n_dst, int heads,
+                          float slope, float* alpha) {
+    if (n_dst == 0) return;
+    gat_alpha_fwd_kernel<<<grid_for(n_dst * heads, BLOCK), BLOCK, 0, s>>>(
+        asrc, adst, src, dst_ptr, n_dst, heads, slope, alpha);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+void launch_gat_alpha_bwd(hipStream_t s, const float* grad_alpha,
+                          const float* alpha, const float* asrc,
+                          const float* adst, const int64_t* src,
+                          const int64_t* dst_ptr, int64_t n_dst, int heads,
+                          float slope, float* g_asrc, float* g_adst) {
+    if (n_dst == 0) return;
+    gat_alpha_bwd_kernel<<<grid_for(n_dst * heads, BLOCK), BLOCK, 0, s>>>(
+        grad_alpha, alpha, asrc, adst, src, dst_ptr, n_dst, heads, slope,
+        g_asrc, g_adst);
     QK_CHECK_HIP(hipGetLastError());
 }
 

@@ -69,6 +69,28 @@ class _SegmentWSum(torch.autograd.Function):
                 gw if ctx.needs_input_grad[1] else None, None, None, None)
 
 
+class _GatAlpha(torch.autograd.Function):
+    """Fused GAT attention coefficients over dst-sorted edges: per-node
+    logit gathers + add + leaky_relu + segment softmax in one kernel pair
+    (no [E,H] torch intermediates)."""
+
+    @staticmethod
+    def forward(ctx, asrc, adst, src, dst_ptr, heads, slope, n_edges):
+        alpha = _ext.gat_alpha(asrc, adst, src, dst_ptr, heads, slope,
+                               n_edges)
+        ctx.save_for_backward(alpha, asrc, adst, src, dst_ptr)
+        ctx.heads, ctx.slope = heads, slope
+        return alpha
+
+    @staticmethod
+    def backward(ctx, grad_alpha):
+        alpha, asrc, adst, src, dst_ptr = ctx.saved_tensors
+        g_asrc, g_adst = _ext.gat_alpha_backward(
+            grad_alpha, alpha, asrc, adst, src, dst_ptr, ctx.heads,
+            ctx.slope)
+        return g_asrc, g_adst, None, None, None, None, None
+
+
 class _SegmentSoftmax(torch.autograd.Function):
     """Softmax over dst-sorted edge segments (csrc/segment_kernels.hip).
     torch's scatter_reduce(amax) path lowers to ~120 rocprim sort kernels
@@ -243,14 +265,15 @@ class GATConv(nn.Module):
             h_dst = self.lin(x_dst).view(-1, H, C)
         alpha_src = (h_src * self.att_src).sum(-1)  # [N_src, H]
         alpha_dst = (h_dst * self.att_dst).sum(-1)  # [N_dst, H]
-        alpha = alpha_src[src] + alpha_dst[dst]     # [E, H]
-        alpha = F.leaky_relu(alpha, self.negative_slope)
-        fused = (self.sorted_dst and alpha.is_cuda
-                 and alpha.dtype == torch.float32 and dst.numel() > 0)
+        fused = (self.sorted_dst and alpha_src.is_cuda
+                 and alpha_src.dtype == torch.float32 and dst.numel() > 0)
         if fused:
             dst_ptr = torch.searchsorted(dst, _arange(n_dst + 1, dst.device))
-            alpha = _SegmentSoftmax.apply(alpha, dst_ptr, H)
+            alpha = _GatAlpha.apply(alpha_src, alpha_dst, src, dst_ptr, H,
+                                    self.negative_slope, src.numel())
         else:
+            alpha = alpha_src[src] + alpha_dst[dst]     # [E, H]
+            alpha = F.leaky_relu(alpha, self.negative_slope)
             # segment softmax over incoming edges of each dst (torch path)
             alpha_max = torch.full((n_dst, H), float("-inf"),
                                    dtype=alpha.dtype, device=alpha.device)
