@@ -105,7 +105,9 @@ def build(verbose=True):
             or any(os.path.getmtime(o) > os.path.getmtime(cpp_test_bin)
                    for o in kernel_objs)):
         cpp_test_obj = os.path.join(BUILD, "test_kernels.o")
-        run(common + ["-c", cpp_test_src, "-o", cpp_test_obj])
+        # -Wno-unused-value: assert-style test code ignores hipFree returns
+        run(common + ["-Wno-unused-value", "-c", cpp_test_src,
+                      "-o", cpp_test_obj])
         run(common + [cpp_test_obj] + kernel_objs + ["-o", cpp_test_bin])
     print(f"built {cpp_test_bin}")
 
