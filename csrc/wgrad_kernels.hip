@@ -5,8 +5,8 @@
 // rocBLAS/Tensile handles this shape with stream-K 32x32 macro-tiles and
 // reaches ~130 GB/s on the bench's layer-1 wgrad (1.1 ms for a 140 MB
 // reduction) — 60x off the HBM3E roof.  This kernel is a plain split-K
-// SGEMM: each block owns a K-chunk and a 128x64 C tile, stages 16-row
-// slices of A and B through LDS, accumulates 8x4 per thread in VGPRs and
+// SGEMM: each block owns a K-chunk and a 64x64 C tile, stages 16-row
+// slices of A and B through LDS, accumulates 4x4 per thread in VGPRs and
 // atomically adds its partial into C.  fp32 global atomics make the
 // reduction order nondeterministic (like Tensile's GSU path); tests
 // compare with a K-scaled tolerance.
@@ -24,12 +24,10 @@ namespace qk {
 
 namespace {
 
-constexpr int BM = 128;
+constexpr int BM = 64;
 constexpr int BN = 64;
 constexpr int BK = 16;
-constexpr int TDIM = 16;  // 16x16 threads, 8x4 microtile each
-constexpr int TM = BM / TDIM;  // 8
-constexpr int TN = BN / TDIM;  // 4
+constexpr int TDIM = 16;  // 16x16 threads, 4x4 microtile each
 
 __global__ void __launch_bounds__(TDIM * TDIM)
 wgrad_kernel(const float* __restrict__ A, const float* __restrict__ B,
@@ -47,8 +45,8 @@ wgrad_kernel(const float* __restrict__ A, const float* __restrict__ B,
     const int tx = threadIdx.x;  // -> M direction (4 cols of A^T)
     const int ty = threadIdx.y;  // -> N direction
 
-    float acc[TM][TN] = {};
-    float bsum[TM] = {};
+    float acc[4][4] = {};
+    float bsum[4] = {};
     const bool do_bias = (bias_grad != nullptr) && (blockIdx.y == 0);
 
     for (int64_t k0 = k_beg; k0 < k_end; k0 += BK) {
@@ -75,36 +73,36 @@ wgrad_kernel(const float* __restrict__ A, const float* __restrict__ B,
         __syncthreads();
 #pragma unroll
         for (int kk = 0; kk < BK; ++kk) {
-            float a[TM], b[TN];
+            float a[4], b[4];
 #pragma unroll
-            for (int i = 0; i < TM; ++i) a[i] = As[kk][tx * TM + i];
+            for (int i = 0; i < 4; ++i) a[i] = As[kk][tx * 4 + i];
 #pragma unroll
-            for (int j = 0; j < TN; ++j) b[j] = Bs[kk][ty * TN + j];
+            for (int j = 0; j < 4; ++j) b[j] = Bs[kk][ty * 4 + j];
 #pragma unroll
-            for (int i = 0; i < TM; ++i)
+            for (int i = 0; i < 4; ++i)
 #pragma unroll
-                for (int j = 0; j < TN; ++j) acc[i][j] += a[i] * b[j];
+                for (int j = 0; j < 4; ++j) acc[i][j] += a[i] * b[j];
             if (do_bias && ty == 0)
 #pragma unroll
-                for (int i = 0; i < TM; ++i) bsum[i] += a[i];
+                for (int i = 0; i < 4; ++i) bsum[i] += a[i];
         }
         __syncthreads();
     }
 
 #pragma unroll
-    for (int i = 0; i < TM; ++i) {
-        int m = m0 + tx * TM + i;
+    for (int i = 0; i < 4; ++i) {
+        int m = m0 + tx * 4 + i;
         if (m >= M) continue;
 #pragma unroll
-        for (int j = 0; j < TN; ++j) {
-            int n = n0 + ty * TN + j;
+        for (int j = 0; j < 4; ++j) {
+            int n = n0 + ty * 4 + j;
             if (n < N) atomicAdd(&C[(int64_t)m * N + n], acc[i][j]);
         }
     }
     if (do_bias && ty == 0)
 #pragma unroll
-        for (int i = 0; i < TM; ++i) {
-            int m = m0 + tx * TM + i;
+        for (int i = 0; i < 4; ++i) {
+            int m = m0 + tx * 4 + i;
             if (m < M) atomicAdd(&bias_grad[m], bsum[i]);
         }
 }
