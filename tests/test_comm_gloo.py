@@ -68,3 +68,41 @@ def _worker(rank, world, port, n_nodes, dim):
 def test_dist_feature_exchange_gloo():
     port = 29511
     mp.spawn(_worker, args=(2, port, 40, 5), nprocs=2, join=True)
+
+
+def test_schedule_properties_random():
+    """Property test: every positive comm_mat entry between distinct hosts
+    is scheduled exactly once, and within one round each host appears in
+    at most one host-pair."""
+    import random
+    rng = random.Random(0)
+    for trial in range(25):
+        hosts = rng.randint(1, 5)
+        rph = rng.randint(1, 3)
+        t = HostRankTable(hosts, rph)
+        ws = hosts * rph
+        comm_mat = [[rng.choice([0, 0, 1, 7]) if t.host(i) != t.host(j)
+                     else 0 for j in range(ws)] for i in range(ws)]
+        steps = schedule(comm_mat, t)
+        seen = set()
+        for step in steps:
+            hosts_in_step = set()
+            host_pairs = set()
+            for (src, dst) in step:
+                pair = (t.host(src), t.host(dst))
+                host_pairs.add(pair)
+                assert (src, dst) not in seen, "pair scheduled twice"
+                seen.add((src, dst))
+            for hs, hd in host_pairs:
+                # a host participates in at most one pair per round
+                assert sum(1 for p in host_pairs if hs in p) <= 2
+                hosts_in_step.update([hs, hd])
+        # completeness: every requested (src_rank -> peer) with traffic is
+        # scheduled (for the peer the table routes that src to)
+        for src in range(ws):
+            for host in range(hosts):
+                if host == t.host(src):
+                    continue
+                dst = t.remote_peer(src, host)
+                if comm_mat[src][dst] > 0:
+                    assert (src, dst) in seen, (src, dst, comm_mat)
