@@ -78,3 +78,25 @@ def test_serving_gpu_pipeline(tmp_path):
     for r in results:
         assert r.shape == (16, 4)
         assert torch.isfinite(r).all()
+
+
+def test_feature_from_mmap_device_config(tmp_path):
+    """Pre-partitioned load path (reference Feature.from_mmap +
+    DeviceConfig): rows selected from an mmap-ed numpy array by explicit
+    per-device index tensors; local order set afterwards."""
+    n, dim = 2000, 32
+    full = np.random.default_rng(0).standard_normal((n, dim)).astype(
+        np.float32)
+    np.save(tmp_path / "feat.npy", full)
+    mm = np.load(tmp_path / "feat.npy", mmap_mode="r")
+
+    hot = torch.arange(0, 500)          # rows 0..499 on GPU
+    cold = torch.arange(500, 2000)      # rest pinned host
+    feature = quiver.Feature(0, device_list=[0], device_cache_size=0)
+    feature.from_mmap(mm, quiver.DeviceConfig({0: hot}, cold))
+    # local order: feature row i of the store = global id order[i]
+    feature.set_local_order(torch.cat([hot, cold]))
+
+    idx = torch.randint(0, n, (777,))
+    got = feature[idx.cuda()].cpu()
+    assert torch.allclose(got, torch.from_numpy(full[idx.numpy()]))
