@@ -90,8 +90,9 @@ def test_feature_from_mmap_device_config(tmp_path):
     np.save(tmp_path / "feat.npy", full)
     mm = np.load(tmp_path / "feat.npy", mmap_mode="r")
 
-    hot = torch.arange(0, 500)          # rows 0..499 on GPU
-    cold = torch.arange(500, 2000)      # rest pinned host
+    # non-identity placement: middle rows hot on GPU, the rest pinned host
+    hot = torch.arange(1000, 1500)
+    cold = torch.cat([torch.arange(0, 1000), torch.arange(1500, 2000)])
     feature = quiver.Feature(0, device_list=[0], device_cache_size=0)
     feature.from_mmap(mm, quiver.DeviceConfig({0: hot}, cold))
     # local order: feature row i of the store = global id order[i]
