@@ -154,9 +154,10 @@ class Topo:
 def reindex_by_config(adj_csr: CSRTopo, graph_feature, gpu_portion: float):
     """Degree-sorted hot/cold ordering (reference utils.py:229-241).
 
-    Returns (reordered_feature, prev_order) where prev_order[new_row] =
-    original node id.  The hottest `gpu_portion` of nodes is partially
-    shuffled so p2p-sharded caches get balanced access frequency.
+    Returns (reordered_feature, new_order) where new_order[original_id]
+    = its row in the reordered feature (the remap used at lookup time).
+    The hottest `gpu_portion` of nodes is shuffled among themselves so
+    p2p-sharded caches get balanced access frequency.
     """
     node_count = adj_csr.node_count
     total_range = torch.arange(node_count, dtype=torch.long)

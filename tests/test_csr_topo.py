@@ -84,3 +84,25 @@ def test_topo_clique_discovery(monkeypatch):
     assert t2.p2p_clique(0) == [0, 1]
     assert t2.p2p_clique(2) == [2, 3]
     assert t2.get_clique_id(1) == 0 and t2.get_clique_id(3) == 1
+
+
+def test_reindex_by_config_invariants():
+    from quiver.utils import reindex_by_config
+    g = torch.Generator().manual_seed(5)
+    n = 400
+    src = torch.randint(0, n, (6000,), generator=g)
+    dst = torch.randint(0, n, (6000,), generator=g)
+    topo = quiver.CSRTopo(torch.stack([src, dst]), node_count=n)
+    feat = torch.randn(n, 8, generator=g)
+    torch.manual_seed(0)
+    reordered, new_order = reindex_by_config(topo, feat.clone(), 0.25)
+    # lookup invariant: reordered[new_order[v]] == feat[v]
+    assert torch.equal(reordered[new_order], feat)
+    # the first 25% of rows hold (a shuffle of) the 25% highest-degree nodes
+    deg = topo.indptr[1:] - topo.indptr[:-1]
+    hot_n = int(n * 0.25)
+    hot_rows_global = set(torch.nonzero(new_order < hot_n).flatten()
+                          .tolist())
+    top_by_degree = set(torch.sort(deg, descending=True)[1][:hot_n].tolist())
+    # degree ties at the boundary allow some slack
+    assert len(hot_rows_global & top_by_degree) >= hot_n * 0.8
