@@ -79,3 +79,35 @@ def test_auto_routing_threshold(tmp_path):
     assert list(heavy) == [0, 1]
     assert list(light) == [50, 51]
     batcher.stop()
+
+
+def test_auto_despatch_routing(tmp_path):
+    """Threshold routing: heavy requests -> GPU queue, light -> CPU queue.
+    Runs the despatcher body synchronously (no worker procs)."""
+    import numpy as np
+    nn = np.zeros(100, dtype=np.int64)
+    nn[50:] = 1000  # nodes >= 50 are predicted-heavy
+    path = tmp_path / "nn.npy"
+    np.save(path, nn)
+
+    batcher = quiver.RequestBatcher(
+        device_num=1, stream_queue_list=[mp.get_context("spawn").Queue()],
+        input_proc_per_device=0, sample_mode="Auto", request_mode="Serve",
+        threshold=1500, neighbour_path=str(path))
+    sq = batcher.stream_queue_list[0]
+    light = np.array([1, 2, 3])            # work 0
+    heavy = np.array([60, 61])             # work 2000 > 1500
+    sq.put(light)
+    sq.put(heavy)
+    from quiver.serving import _Stop
+    sq.put(_Stop())
+    batcher.auto_despatch(0)               # runs to the sentinel
+
+    cpu_q = batcher.cpu_batched_queue_list[0]
+    gpu_q = batcher.gpu_batched_queue_list[0]
+    got_cpu = cpu_q.get(timeout=10)
+    got_gpu = gpu_q.get(timeout=10)
+    assert list(got_cpu) == [1, 2, 3]
+    assert list(got_gpu) == [60, 61]
+    assert isinstance(gpu_q.get(timeout=10), _Stop)
+    assert isinstance(cpu_q.get(timeout=10), _Stop)
