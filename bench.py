@@ -91,8 +91,13 @@ def main():
     p.add_argument("--preset", default="products", choices=sorted(PRESETS))
     p.add_argument("--cache", default=None,
                    help="per-GPU HBM feature cache (default: preset's)")
-    p.add_argument("--cache-policy", default="device_replicate",
-                   choices=["device_replicate", "p2p_clique_replicate"])
+    p.add_argument("--cache-policy", default=None,
+                   choices=["device_replicate", "p2p_clique_replicate"],
+                   help="default: device_replicate for 1 GPU (the reference's"
+                        " headline 1-GPU config); p2p_clique_replicate for"
+                        " N>1 (BASELINE config 3: feature store sharded +"
+                        " replicated over xGMI — 8 x 20%% cache shards hold"
+                        " the whole products feature tensor in HBM)")
     p.add_argument("--nodes", type=int, default=None)
     p.add_argument("--edges", type=int, default=None)
     p.add_argument("--no-overlap", action="store_true",
@@ -128,6 +133,9 @@ def main():
                          f"available {avail/2**30:.0f} GiB")
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
+    if args.cache_policy is None:
+        args.cache_policy = ("p2p_clique_replicate" if world > 1
+                             else "device_replicate")
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     assert world == args.gpus or world == 1, \
