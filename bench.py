@@ -158,18 +158,30 @@ def main():
     for beg in range(0, args.nodes, step):
         feat_cpu[beg:beg + step].uniform_(-1.0, 1.0, generator=g)
 
-    if args.cache_policy == "p2p_clique_replicate":
-        quiver.init_p2p(list(range(world)))
-        device_list = list(range(world))
-    else:
-        device_list = [local_rank]
+    def build_feature(policy):
+        if policy == "p2p_clique_replicate":
+            quiver.init_p2p(list(range(world)))
+            devices = list(range(world))
+        else:
+            devices = [local_rank]
+        f = quiver.Feature(local_rank, device_list=devices,
+                           device_cache_size=args.cache,
+                           cache_policy=policy, csr_topo=csr_topo)
+        f.from_cpu_tensor(feat_cpu)
+        return f
+
     sampler = quiver.GraphSageSampler(csr_topo, FANOUT, device=local_rank,
                                       mode=args.mode)
-    feature = quiver.Feature(local_rank, device_list=device_list,
-                             device_cache_size=args.cache,
-                             cache_policy=args.cache_policy,
-                             csr_topo=csr_topo)
-    feature.from_cpu_tensor(feat_cpu)
+    try:
+        feature = build_feature(args.cache_policy)
+    except (RuntimeError, AssertionError) as e:
+        if args.cache_policy != "p2p_clique_replicate":
+            raise
+        # unvalidated multi-process xGMI layout must not zero the run
+        print(f"# rank {rank}: p2p_clique_replicate failed ({e}); "
+              "falling back to device_replicate", flush=True)
+        args.cache_policy = "device_replicate"
+        feature = build_feature(args.cache_policy)
 
     if args.model == "gat":
         model = GAT(feat_dim, HIDDEN // 4, n_classes,
