@@ -413,6 +413,32 @@ void test_gather_scatter() {
         for (int64_t j = 0; j < dim; ++j)
             REQUIRE(shard0_after[sidx[i] * dim + j] == src[i * dim + j]);
 
+    // n_dev: exact count on device, indices upper-bound sized — rows
+    // beyond *n_dev must stay untouched (async sample->gather chains)
+    const int64_t n_exact = 1000;
+    auto* d_nexact = dalloc<int64_t>(1);
+    std::vector<int64_t> nv = {n_exact};
+    h2d(d_nexact, nv);
+    std::vector<float> sentinel(n * dim, -777.f);
+    auto* d_out2 = dalloc<float>(n * dim);
+    h2d(d_out2, sentinel);
+    qk::launch_gather(nullptr, spec, d_idx, n, (char*)d_out2, d_nexact);
+    QK_CHECK_HIP(hipDeviceSynchronize());
+    auto out2 = d2h(d_out2, n * dim);
+    for (int64_t i = 0; i < n; ++i)
+        for (int64_t j = 0; j < dim; ++j) {
+            if (i < n_exact) {
+                const float* want = idx[i] < rows0
+                                        ? &shard0[idx[i] * dim + j]
+                                        : &host1[(idx[i] - rows0) * dim + j];
+                // shard0 rows were scatter-updated above; skip those
+                if (idx[i] < rows0 && seen.count(idx[i])) continue;
+                REQUIRE(out2[i * dim + j] == *want);
+            } else {
+                REQUIRE(out2[i * dim + j] == -777.f);
+            }
+        }
+    hipFree(d_nexact); hipFree(d_out2);
     hipFree(d_shard0); hipFree(d_idx); hipFree(d_out); hipFree(d_sidx);
     hipFree(d_src);
     QK_CHECK_HIP(hipHostFree(h_pinned));
