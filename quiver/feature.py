@@ -103,6 +103,14 @@ class Feature(object):
             for clique_id, clique_devices in enumerate(self.topo.cliques):
                 block_size = self.cal_size(
                     cpu_tensor, cache_memory_budget // len(clique_devices))
+                # when the aggregate clique budget exceeds the tensor
+                # (e.g. 8 x 20%), balance shards across all devices
+                # instead of filling the first few and leaving the rest
+                # empty — xGMI read traffic then spreads over every link
+                block_size = min(block_size,
+                                 (cache_part.shape[0] +
+                                  len(clique_devices) - 1)
+                                 // len(clique_devices))
                 shard_tensor = ShardTensor(self.rank, ShardTensorConfig({}))
                 cur_pos = 0
                 for idx, device in enumerate(clique_devices):
