@@ -44,7 +44,7 @@ def test_cpu_training_learns():
             opt.zero_grad()
             loss.backward()
             opt.step()
-            total += float(loss)
+            total += float(loss.detach())
         if first_loss is None:
             first_loss = total
     assert total < first_loss * 0.8, (first_loss, total)

@@ -42,7 +42,7 @@ def test_training_step(training_setup, mode):
         opt.zero_grad()
         loss.backward()
         opt.step()
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     assert all(torch.isfinite(torch.tensor(losses)))
 
 
