@@ -56,6 +56,57 @@ struct DeviceScope {
     ~DeviceScope() { (void)hipSetDevice(prev); }
 };
 
+// Chunked hipHostRegister for multi-GB UVA buffers (papers100M indices are
+// 12.5 GB): registering in 1 GB slices bounds the per-call pinning peak —
+// the reference's hard-won lesson (quiverRegister, quiver.cu.hpp:16-26).
+// Kernels need ONE contiguous device view, so the per-chunk device
+// pointers are verified contiguous; if the runtime ever maps them apart,
+// everything is unregistered and one whole-buffer registration is used
+// instead.  Returns the device-visible base pointer; appends every
+// registered host pointer to `registered` (for hipHostUnregister later).
+// Already-registered memory (e.g. a shared tensor registered by a prior
+// ShardTensor) is detected on the first chunk and reused.
+void* register_host_chunked(void* p, size_t bytes,
+                            std::vector<void*>& registered) {
+    constexpr size_t kChunk = 1ull << 30;
+    constexpr unsigned kFlags =
+        hipHostRegisterMapped | hipHostRegisterPortable;
+    auto devptr = [](void* hp) {
+        void* dp = nullptr;
+        QK_CHECK_HIP(hipHostGetDevicePointer(&dp, hp, 0));
+        return dp;
+    };
+    char* base = (char*)p;
+    std::vector<void*> regs;
+    void* dp0 = nullptr;
+    bool contiguous = true;
+    for (size_t off = 0; off < bytes; off += kChunk) {
+        size_t len = std::min(kChunk, bytes - off);
+        hipError_t err = hipHostRegister(base + off, len, kFlags);
+        if (err == hipErrorHostMemoryAlreadyRegistered && off == 0) {
+            (void)hipGetLastError();
+            return devptr(base);  // whole buffer registered by someone else
+        }
+        if (err != hipSuccess) {
+            for (void* r : regs) (void)hipHostUnregister(r);
+            (void)hipGetLastError();
+            QK_CHECK_HIP(err);  // throws with the error string
+        }
+        regs.push_back(base + off);
+        void* dp = devptr(base + off);
+        if (off == 0) dp0 = dp;
+        else if ((char*)dp != (char*)dp0 + off) contiguous = false;
+    }
+    if (!contiguous) {
+        for (void* r : regs) (void)hipHostUnregister(r);
+        QK_CHECK_HIP(hipHostRegister(p, bytes, kFlags));
+        registered.push_back(p);
+        return devptr(p);
+    }
+    registered.insert(registered.end(), regs.begin(), regs.end());
+    return dp0;
+}
+
 // Persistent worker pool for the CPU-staged gather.  Spawning threads
 // per call costs ~2 ms (64 x ~30 us) — more than the copy itself — and
 // at::parallel_for from a non-main thread degrades to serial; a resident
@@ -433,12 +484,7 @@ class GpuSampler {
 
   private:
     void* register_host(void* p, size_t bytes) {
-        QK_CHECK_HIP(hipHostRegister(
-            p, bytes, hipHostRegisterMapped | hipHostRegisterPortable));
-        registered_.push_back(p);
-        void* dp = nullptr;
-        QK_CHECK_HIP(hipHostGetDevicePointer(&dp, p, 0));
-        return dp;
+        return register_host_chunked(p, bytes, registered_);
     }
 
     int device_;
@@ -577,7 +623,7 @@ struct ShardItem {
     bool owned_hip = false;   // free with hipFree
     bool from_ipc = false;    // close with hipIpcCloseMemHandle
     torch::Tensor keeper;     // keeps host/tensor storage alive
-    void* host_reg = nullptr; // hipHostUnregister on destruction
+    std::vector<void*> host_regs;  // hipHostUnregister on destruction
 };
 
 class ShardTensorItem {
@@ -616,7 +662,7 @@ class ShardTensor {
                 DeviceScope g(device_);
                 (void)hipIpcCloseMemHandle(s.dptr);
             }
-            if (s.host_reg) (void)hipHostUnregister(s.host_reg);
+            for (void* r : s.host_regs) (void)hipHostUnregister(r);
         }
     }
 
@@ -645,19 +691,10 @@ class ShardTensor {
             TORCH_CHECK(tensor.device().is_cpu(),
                         "host shard expects a CPU tensor");
             // zero-copy pinned host tier: register the tensor's own memory
-            // (shared-memory friendly) and read it from gather kernels.
-            hipError_t err = hipHostRegister(
-                tensor.data_ptr(), bytes,
-                hipHostRegisterMapped | hipHostRegisterPortable);
-            if (err == hipErrorHostMemoryAlreadyRegistered) {
-                (void)hipGetLastError();
-            } else {
-                QK_CHECK_HIP(err);
-                item.host_reg = tensor.data_ptr();
-            }
-            void* dp = nullptr;
-            QK_CHECK_HIP(hipHostGetDevicePointer(&dp, tensor.data_ptr(), 0));
-            item.dptr = dp;
+            // (shared-memory friendly, 1 GB chunks) and read it from
+            // gather kernels.
+            item.dptr = register_host_chunked(tensor.data_ptr(), bytes,
+                                              item.host_regs);
             item.keeper = tensor;
         }
         shards_.push_back(std::move(item));

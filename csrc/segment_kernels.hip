@@ -326,8 +326,10 @@ void launch_segment_mean_bwd(hipStream_t s, const float* grad_out,
     QK_CHECK_HIP(hipGetLastError());
 }
 
-static void check_chead(int chead) {
-    if (chead % VPL != 0)
+static void check_chead(int heads, int chead) {
+    // heads == 1: the whole dim is one head, a vector chunk cannot
+    // straddle a head boundary regardless of chead
+    if (heads > 1 && chead % VPL != 0)
         throw std::runtime_error(
             "segment_wsum: per-head channels must be a multiple of 4 "
             "so vector chunks stay within one head");
@@ -337,7 +339,7 @@ void launch_segment_wsum_fwd(hipStream_t s, const float* x, const float* w,
                              const int64_t* src, const int64_t* dst_ptr,
                              int64_t n_dst, int heads, int chead, float* out) {
     if (n_dst == 0) return;
-    check_chead(chead);
+    check_chead(heads, chead);
     segment_wsum_fwd_kernel<<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0, s>>>(
         x, w, src, dst_ptr, n_dst, heads, chead, out);
     QK_CHECK_HIP(hipGetLastError());
@@ -348,6 +350,7 @@ void launch_segment_wsum_bwd_x(hipStream_t s, const float* grad_out,
                                const int64_t* dst_ptr, int64_t n_dst,
                                int heads, int chead, float* grad_x) {
     if (n_dst == 0) return;
+    check_chead(heads, chead);
     segment_wsum_bwd_x_kernel<<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0,
                                 s>>>(grad_out, w, src, dst_ptr, n_dst, heads,
                                      chead, grad_x);
@@ -359,6 +362,7 @@ void launch_segment_wsum_bwd_w(hipStream_t s, const float* grad_out,
                                const int64_t* dst_ptr, int64_t n_dst,
                                int heads, int chead, float* grad_w) {
     if (n_dst == 0) return;
+    check_chead(heads, chead);
     segment_wsum_bwd_w_kernel<<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0,
                                 s>>>(grad_out, x, src, dst_ptr, n_dst, heads,
                                      chead, grad_w);

@@ -14,34 +14,34 @@ from . import _ext
 
 
 class HostRankTable:
+    """rank <-> host arithmetic for a homogeneous layout: world ranks are
+    numbered host-major, so rank r lives on host r // rank_per_host and a
+    rank's peer on a remote host is the one sharing its local index."""
+
     def __init__(self, hosts, rank_per_host):
         self.hosts = hosts
         self.rank_per_host = rank_per_host
-        self.host2ranks = {}
-        self.rank2host = []
-        cnt = 0
-        for i in range(hosts):
-            self.host2ranks[i] = list(range(cnt, cnt + rank_per_host))
-            cnt += rank_per_host
-            self.rank2host.extend([i] * rank_per_host)
+
+    @property
+    def world(self):
+        return self.hosts * self.rank_per_host
 
     def ranks(self, host):
-        return self.host2ranks[host]
+        base = host * self.rank_per_host
+        return list(range(base, base + self.rank_per_host))
 
     def host(self, rank):
-        return self.rank2host[rank]
+        return rank // self.rank_per_host
 
     def remote_peer(self, rank, host):
-        return self.ranks(host)[rank % self.rank_per_host]
+        return host * self.rank_per_host + rank % self.rank_per_host
 
     def remote_peers(self, rank, hosts):
         return [(rank, self.remote_peer(rank, host)) for host in hosts]
 
     def get_comm_mat(self, flat_allreduce):
-        flat = flat_allreduce.to("cpu")
-        size = self.hosts * self.rank_per_host
-        return [[int(flat[i * size + j]) for j in range(size)]
-                for i in range(size)]
+        w = self.world
+        return flat_allreduce.cpu().reshape(w, w).tolist()
 
 
 def schedule(comm_mat, table):

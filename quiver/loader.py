@@ -49,6 +49,15 @@ class TrainingPrefetcher:
                 return False
             if getattr(self.feature, "mmap_handle_", None) is not None:
                 return False
+            # pre-validate the known unsupported layout (shards needing the
+            # cross-clique pass) instead of learning it from a RuntimeError
+            # on the first produce
+            try:
+                st = self.feature._shard_tensor()
+                if st._inaccessible_ranges():
+                    return False
+            except Exception:
+                pass  # lazy IPC store: decided on first produce instead
         return True
 
     def __iter__(self):
@@ -84,8 +93,16 @@ class TrainingPrefetcher:
                         ev.record(side)
                         pending.append(("tok", tok, x_ub, ev))
                         return True
-                    except RuntimeError:
-                        chain_async[0] = False  # fall through, stay sync
+                    except RuntimeError as e:
+                        # fall through to the sync path for the rest of the
+                        # epoch — but say why, so a transient failure (e.g.
+                        # OOM during gather_raw) is not silently converted
+                        # into a slower epoch
+                        import warnings
+                        warnings.warn(
+                            "TrainingPrefetcher: zero-sync chain disabled "
+                            f"after: {e}", RuntimeWarning)
+                        chain_async[0] = False
                 n_id, bs, adjs = self.sampler.sample(seeds)
                 x = self.feature[n_id] if self.feature is not None else None
                 ev = torch.cuda.Event()

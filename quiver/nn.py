@@ -166,13 +166,27 @@ def _arange(n, device):
     return cached[:n]
 
 
+_QUIVER_DEBUG = bool(int(__import__("os").environ.get("QUIVER_DEBUG", "0")))
+
+
+def _dst_ptr_from_sorted(dst, n_dst):
+    """Segment pointer from a dst-sorted edge list.  searchsorted silently
+    returns garbage on unsorted input, so sorted_dst=True layers fed edges
+    from a non-quiver sampler would be silently wrong — QUIVER_DEBUG=1
+    turns that into a loud error."""
+    if _QUIVER_DEBUG and dst.numel() > 1:
+        assert bool((dst[1:] >= dst[:-1]).all()), \
+            "sorted_dst=True but edge_index[1] is not sorted ascending"
+    return torch.searchsorted(dst, _arange(n_dst + 1, dst.device))
+
+
 def _mean_aggregate(x_src, src, dst, n_dst, sorted_dst=False):
     """Mean of x_src[src] grouped by dst.  Uses the fused HIP kernel when
     the caller guarantees dst is sorted ascending (our sampler's layout)
     on fp32 GPU tensors."""
     if (sorted_dst and x_src.is_cuda and x_src.dtype == torch.float32
             and dst.numel() > 0):
-        dst_ptr = torch.searchsorted(dst, _arange(n_dst + 1, dst.device))
+        dst_ptr = _dst_ptr_from_sorted(dst, n_dst)
         return _SegmentMeanAgg.apply(x_src, src, dst_ptr)
     agg = torch.zeros((n_dst, x_src.size(1)), dtype=x_src.dtype,
                       device=x_src.device)
@@ -257,7 +271,9 @@ class GATConv(nn.Module):
 
         h_src = self.lin(x_src).view(-1, H, C)
         if (x_dst.data_ptr() == x_src.data_ptr()
-                and x_dst.size(0) <= x_src.size(0)):
+                and x_dst.size(0) <= x_src.size(0)
+                and x_dst.stride() == x_src.stride()
+                and x_dst.size(1) == x_src.size(1)):
             # bipartite prefix convention: x_dst is x_src[:n_dst] — reuse
             # the projection instead of a second GEMM (+ its backward)
             h_dst = h_src[:x_dst.size(0)]
@@ -268,7 +284,7 @@ class GATConv(nn.Module):
         fused = (self.sorted_dst and alpha_src.is_cuda
                  and alpha_src.dtype == torch.float32 and dst.numel() > 0)
         if fused:
-            dst_ptr = torch.searchsorted(dst, _arange(n_dst + 1, dst.device))
+            dst_ptr = _dst_ptr_from_sorted(dst, n_dst)
             alpha = _GatAlpha.apply(alpha_src, alpha_dst, src, dst_ptr, H,
                                     self.negative_slope, src.numel())
         else:
@@ -293,8 +309,7 @@ class GATConv(nn.Module):
                 and dst.numel() > 0):
             # fused weighted segment sum over the dst-sorted edges
             if not fused:
-                dst_ptr = torch.searchsorted(dst,
-                                             _arange(n_dst + 1, dst.device))
+                dst_ptr = _dst_ptr_from_sorted(dst, n_dst)
             out = _SegmentWSum.apply(h_src.reshape(-1, H * C), alpha, src,
                                      dst_ptr, H)
             out = out if self.concat else out.view(n_dst, H, C).mean(1)

@@ -206,7 +206,9 @@ def _drain_after_stop(q, stop_item):
         except _queue.Empty:
             break
         if isinstance(item, _Stop):
-            q.put(item)
+            # consume (don't re-queue) sentinels while draining: re-putting
+            # on every sighting made extra sentinels accumulate in the
+            # shared queue.  Exactly one goes back at the end.
             stops += 1
             time.sleep(0.05)
             continue

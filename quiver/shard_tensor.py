@@ -4,7 +4,7 @@
 Capability parity with reference quiver/shard_tensor.py; the device kernel
 is csrc/gather_kernels.hip.
 """
-from typing import Dict
+from typing import Dict, NamedTuple
 
 import torch
 
@@ -12,32 +12,11 @@ from . import _ext
 from .utils import parse_size
 
 
-class Offset:
-    def __init__(self, start, end):
-        self.start_ = start
-        self.end_ = end
-
-    @property
-    def start(self):
-        return self.start_
-
-    @property
-    def end(self):
-        return self.end_
-
-
-class DeviceCollectionJob:
-    def __init__(self, part_orders, request_nodes):
-        self.part_orders_ = part_orders
-        self.request_nodes_ = request_nodes
-
-    @property
-    def part_orders(self):
-        return self.part_orders_
-
-    @property
-    def request_nodes(self):
-        return self.request_nodes_
+class Offset(NamedTuple):
+    """Row span [start, end) a device's shard occupies in the virtual
+    tensor (named for reference-API familiarity; it is just a row range)."""
+    start: int
+    end: int
 
 
 class ShardTensorConfig:
