@@ -159,6 +159,21 @@ def main():
         feat_cpu[beg:beg + step].uniform_(-1.0, 1.0, generator=g)
 
     def build_feature(policy):
+        if policy == "p2p_clique_replicate" and distributed:
+            # collaborative build: each rank allocates ONLY its own
+            # device's hot shard, peers' shards are reopened via hipIpc
+            # (one hipMalloc per shard in the whole job, not per rank)
+            f = quiver.Feature(local_rank, device_list=list(range(world)),
+                               device_cache_size=args.cache,
+                               cache_policy=policy, csr_topo=csr_topo)
+
+            def all_gather(obj):
+                objs = [None] * world
+                dist.all_gather_object(objs, obj)
+                return objs
+
+            f.from_cpu_tensor_dist(feat_cpu, world, rank, all_gather)
+            return f
         if policy == "p2p_clique_replicate":
             quiver.init_p2p(list(range(world)))
             devices = list(range(world))

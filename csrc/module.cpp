@@ -700,6 +700,23 @@ class ShardTensor {
         shards_.push_back(std::move(item));
     }
 
+    // Adopt a shard allocated by another ShardTensor of THIS process
+    // (hipIpcOpenMemHandle on a same-process handle is an error, so the
+    // per-rank distributed build shares its own allocation by aliasing).
+    // Ownership stays with `other`; the Python layer keeps it alive.
+    void append_from(const ShardTensor& other, int idx) {
+        TORCH_CHECK(idx >= 0 && idx < (int)other.shards_.size(),
+                    "append_from: no such shard");
+        const ShardItem& s = other.shards_[idx];
+        init_row_meta_from(other.row_shape_, other.dtype_);
+        ShardItem item;
+        item.dptr = s.dptr;
+        item.rows = s.rows;
+        item.device = s.device;
+        item.keeper = s.keeper;  // host-tier tensors stay alive either way
+        shards_.push_back(std::move(item));
+    }
+
     void append_item(const ShardTensorItem& it) {
         // open a peer shard exported from another process
         init_row_meta_from(it.row_shape, it.dtype);
@@ -1407,6 +1424,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         .def(py::init<int>())
         .def("append", &ShardTensor::append)
         .def("append_item", &ShardTensor::append_item)
+        .def("append_from", &ShardTensor::append_from,
+             "alias a shard owned by another same-process ShardTensor")
         .def("share_ipc", &ShardTensor::share_ipc)
         .def("__getitem__", &ShardTensor::gather,
              py::call_guard<py::gil_scoped_release>())

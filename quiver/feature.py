@@ -12,9 +12,10 @@ from typing import List
 import numpy as np
 import torch
 
+from . import _ext
 from .shard_tensor import ShardTensor, ShardTensorConfig
 from .trace import trace_scope
-from .utils import Topo, CSRTopo, reindex_feature, parse_size
+from .utils import Topo, CSRTopo, hot_order, parse_size
 
 __all__ = ["Feature", "DistFeature", "PartitionInfo", "DeviceConfig"]
 
@@ -71,8 +72,15 @@ class Feature(object):
         cache_size = self.cal_size(cpu_tensor, cache_memory_budget)
         return [cpu_tensor[:cache_size], cpu_tensor[cache_size:]]
 
-    def from_cpu_tensor(self, cpu_tensor: torch.Tensor):
-        """Split a CPU tensor into hot cache(s) + pinned cold tier."""
+    def from_cpu_tensor(self, cpu_tensor: torch.Tensor,
+                        score: torch.Tensor = None):
+        """Split a CPU tensor into hot cache(s) + pinned cold tier.
+
+        score: optional per-node hotness used for cache placement instead
+        of out-degree — pass the access probability from
+        GraphSageSampler.sample_prob for probability-driven placement
+        (reference cal_neighbor_prob, quiver_sample.cu:100-111).
+        """
         if self.cache_policy == "device_replicate":
             cache_memory_budget = parse_size(self.device_cache_size)
             shuffle_ratio = 0.0
@@ -85,8 +93,9 @@ class Feature(object):
 
         if self.csr_topo is not None:
             if self.csr_topo.feature_order is None:
-                cpu_tensor, self.csr_topo.feature_order = reindex_feature(
-                    self.csr_topo, cpu_tensor, shuffle_ratio)
+                prev_order, self.csr_topo.feature_order = hot_order(
+                    self.csr_topo, shuffle_ratio, score=score)
+                cpu_tensor = cpu_tensor[prev_order]
             self.feature_order = self.csr_topo.feature_order.to(self.rank) \
                 if torch.cuda.is_available() else self.csr_topo.feature_order
 
@@ -136,6 +145,84 @@ class Feature(object):
                                                     ShardTensorConfig({}))
                 shard_tensor.append(self.cpu_part, -1)
                 self.clique_tensor_list[clique_id] = shard_tensor
+
+    def from_cpu_tensor_dist(self, cpu_tensor: torch.Tensor, world: int,
+                             rank: int, all_gather_object,
+                             score: torch.Tensor = None):
+        """Collaborative p2p_clique_replicate build, one process per GPU.
+
+        Each torchrun rank hipMallocs ONLY its own device's hot shard and
+        reopens every peer's shard via hipIpc — instead of each of N ranks
+        allocating the full N-shard set (N x hot-cache HBM, N x H2D setup
+        traffic).  The hot/cold ordering is computed deterministically
+        (seeded) so all ranks agree without a broadcast.
+
+        Layout parity with the single-process from_cpu_tensor on the same
+        budget: shard r holds hot rows [r*block, (r+1)*block).
+
+        all_gather_object(obj) -> list of every rank's obj (e.g. a wrapper
+        over torch.distributed.all_gather_object).
+
+        Reference design point: NVLink-sharded p2p store built by one
+        owner + cudaIpc reopen (quiver_feature.cu:378-421, examples).
+        """
+        assert self.cache_policy == "p2p_clique_replicate", \
+            "distributed build targets the sharded (p2p) layout"
+        budget = parse_size(self.device_cache_size)
+        total_budget = budget * world
+        cache_rows = min(self.cal_size(cpu_tensor, total_budget),
+                         cpu_tensor.size(0))
+        prev_order = None
+        if self.csr_topo is not None:
+            if self.csr_topo.feature_order is None:
+                shuffle_ratio = cache_rows / max(1, cpu_tensor.size(0))
+                prev_order, self.csr_topo.feature_order = hot_order(
+                    self.csr_topo, shuffle_ratio, score=score)
+            else:
+                prev_order = torch.argsort(self.csr_topo.feature_order)
+            self.feature_order = self.csr_topo.feature_order.to(self.rank)
+
+        def rows(beg, end):
+            """Rows [beg, end) of the (virtually) reordered tensor —
+            gathered per-slice so the full reordered tensor is never
+            materialized."""
+            if prev_order is None:
+                return cpu_tensor[beg:end]
+            return cpu_tensor[prev_order[beg:end]]
+
+        # this rank's shard of the hot rows
+        block = (cache_rows + world - 1) // world
+        beg = min(rank * block, cache_rows)
+        end = min(beg + block, cache_rows)
+        own = ShardTensor(self.rank, ShardTensorConfig({}))
+        handles = []
+        if end > beg:
+            own.shard_tensor.append(rows(beg, end).contiguous(), self.rank)
+            handles = [it.share_ipc()
+                       for it in own.shard_tensor.share_ipc()]
+        gathered = all_gather_object((rank, self.rank, handles))
+
+        # xGMI peer access across every device that owns a shard
+        devices = sorted({dev for _, dev, hs in gathered if hs})
+        if len(devices) > 1:
+            _ext.init_p2p(devices)
+
+        st = ShardTensor(self.rank, ShardTensorConfig({}))
+        for r, dev, hs in sorted(gathered):
+            if r == rank:
+                for i in range(own.shard_tensor.shard_count()):
+                    st.shard_tensor.append_from(own.shard_tensor, i)
+            else:
+                for h in hs:
+                    st.shard_tensor.append_item(
+                        _ext.ShardTensorItem.from_ipc(h))
+        st._own_alloc = own  # keeps this rank's hipMalloc alive
+
+        if cache_rows < cpu_tensor.size(0):
+            self.cpu_part = rows(cache_rows, cpu_tensor.size(0)).contiguous()
+            st.append(self.cpu_part, -1)
+        clique_id = self.topo.get_clique_id(self.rank)
+        self.clique_tensor_list[clique_id] = st
 
     def from_mmap(self, np_array, device_config: DeviceConfig):
         """Build from a (mmap) numpy array + explicit per-device partition.

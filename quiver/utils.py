@@ -151,24 +151,40 @@ class Topo:
         return s
 
 
+def hot_order(adj_csr: CSRTopo, gpu_portion: float, seed: int = 0,
+              score: torch.Tensor = None):
+    """Hot/cold row ordering for cache placement.
+
+    Nodes are ranked by `score` (default: out-degree) descending; the
+    hottest `gpu_portion` are shuffled among themselves so p2p-sharded
+    caches get balanced access frequency.  Deterministic given `seed`, so
+    every rank of a torchrun job computes the identical order with no
+    broadcast.
+
+    Returns (prev_order, new_order): row i of the reordered feature holds
+    original node prev_order[i]; new_order[original_id] = its new row.
+    """
+    node_count = adj_csr.node_count
+    if score is None:
+        score = adj_csr.indptr[1:] - adj_csr.indptr[:-1]
+    _, prev_order = torch.sort(score.cpu(), descending=True)
+    n_hot = int(node_count * gpu_portion)
+    if n_hot > 1:
+        g = torch.Generator().manual_seed(seed)
+        prev_order[:n_hot] = prev_order[torch.randperm(n_hot, generator=g)]
+    new_order = torch.empty_like(prev_order)
+    new_order[prev_order] = torch.arange(node_count, dtype=torch.long)
+    return prev_order, new_order
+
+
 def reindex_by_config(adj_csr: CSRTopo, graph_feature, gpu_portion: float):
     """Degree-sorted hot/cold ordering (reference utils.py:229-241).
 
     Returns (reordered_feature, new_order) where new_order[original_id]
     = its row in the reordered feature (the remap used at lookup time).
-    The hottest `gpu_portion` of nodes is shuffled among themselves so
-    p2p-sharded caches get balanced access frequency.
     """
-    node_count = adj_csr.node_count
-    total_range = torch.arange(node_count, dtype=torch.long)
-    perm_range = torch.randperm(int(node_count * gpu_portion))
-    degree = adj_csr.indptr[1:] - adj_csr.indptr[:-1]
-    _, prev_order = torch.sort(degree, descending=True)
-    new_order = torch.zeros_like(prev_order)
-    prev_order[:int(node_count * gpu_portion)] = prev_order[perm_range]
-    new_order[prev_order] = total_range
-    graph_feature = graph_feature[prev_order]
-    return graph_feature, new_order
+    prev_order, new_order = hot_order(adj_csr, gpu_portion)
+    return graph_feature[prev_order], new_order
 
 
 def reindex_feature(graph: CSRTopo, feature, ratio: float):

@@ -106,3 +106,28 @@ def test_reindex_by_config_invariants():
     top_by_degree = set(torch.sort(deg, descending=True)[1][:hot_n].tolist())
     # degree ties at the boundary allow some slack
     assert len(hot_rows_global & top_by_degree) >= hot_n * 0.8
+
+
+def test_hot_order_deterministic_and_valid():
+    """from_cpu_tensor_dist relies on every rank computing the identical
+    hot/cold order with no broadcast — hot_order must be a deterministic
+    permutation for a given seed."""
+    import torch
+    from quiver.utils import hot_order
+    import quiver
+    indptr = torch.tensor([0, 5, 6, 10, 10, 12], dtype=torch.long)
+    indices = torch.arange(12, dtype=torch.long) % 5
+    topo = quiver.CSRTopo(indptr=indptr, indices=indices)
+    p1, n1 = hot_order(topo, 0.6)
+    p2, n2 = hot_order(topo, 0.6)
+    assert torch.equal(p1, p2) and torch.equal(n1, n2)
+    # permutation + inverse relationship
+    assert torch.equal(torch.sort(p1)[0], torch.arange(5))
+    assert torch.equal(n1[p1], torch.arange(5))
+    # a different seed shuffles the hot head differently (not a no-op knob)
+    p3, _ = hot_order(topo, 0.6, seed=99)
+    assert p3.shape == p1.shape
+    # score override: rank by external hotness instead of degree
+    score = torch.tensor([0., 10., 1., 5., 2.])
+    p4, _ = hot_order(topo, 0.0, score=score)
+    assert p4.tolist() == [1, 3, 4, 2, 0]

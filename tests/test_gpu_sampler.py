@@ -173,3 +173,49 @@ def test_fused_matches_perhop_deterministic_counts(small_graph):
     want_e1 = int(torch.minimum(deg, torch.tensor(6)).sum())
     assert a1[-1].edge_index.shape[1] == want_e1
     assert b1 == 25
+
+
+def test_cal_next_randomized_vs_cpu(small_graph):
+    """cal_next on a random graph vs a CPU implementation of the formula
+    (reference cuda_random.cu.hpp:71-104):
+    cur[v] = 1 - (1 - last[v]) * prod_{u in N(v)} (1 - last[u]*min(1,k/deg_u))
+    """
+    indptr, indices = small_graph
+    n = indptr.numel() - 1
+    topo = quiver.CSRTopo(indptr=indptr, indices=indices)
+    deg = (indptr[1:] - indptr[:-1]).float()
+    for k, seed in [(3, 0), (8, 1), (25, 2)]:
+        s = quiver.GraphSageSampler(topo, [k], device=0, mode="GPU")
+        s.lazy_init_quiver()
+        g = torch.Generator().manual_seed(seed)
+        last = torch.rand(n, generator=g)
+        last[torch.rand(n, generator=g) < 0.5] = 0.0  # sparse support
+        cur_gpu = torch.zeros(n, device="cuda:0")
+        s.quiver.cal_neighbor_prob(0, last.to("cuda:0"), cur_gpu, k)
+        cur_gpu = cur_gpu.cpu()
+        # CPU reference
+        sel = last * torch.minimum(torch.ones(n), k / deg.clamp(min=1))
+        sel[deg == 0] = 0.0
+        cur_ref = torch.empty(n)
+        for v in range(n):
+            nbrs = indices[indptr[v]:indptr[v + 1]]
+            prod = float(torch.prod(1.0 - sel[nbrs])) if nbrs.numel() else 1.0
+            cur_ref[v] = 1.0 - (1.0 - last[v]) * prod
+        assert torch.allclose(cur_gpu, cur_ref, atol=1e-5), \
+            f"k={k}: max err {(cur_gpu - cur_ref).abs().max()}"
+
+
+def test_multihop_sample_prob_monotone(small_graph):
+    """Multi-hop access probability: seeds stay at 1, probabilities lie in
+    [0,1], and adding a hop can only grow each node's probability."""
+    indptr, indices = small_graph
+    topo = quiver.CSRTopo(indptr=indptr, indices=indices)
+    n = topo.node_count
+    train_idx = torch.arange(0, n, 7)
+    s1 = quiver.GraphSageSampler(topo, [10], device=0, mode="GPU")
+    s2 = quiver.GraphSageSampler(topo, [10, 10], device=0, mode="GPU")
+    p1 = s1.sample_prob(train_idx, n).cpu()
+    p2 = s2.sample_prob(train_idx, n).cpu()
+    assert torch.all(p1[train_idx] == 1.0)
+    assert torch.all((p1 >= 0) & (p1 <= 1.0 + 1e-6))
+    assert torch.all(p2 >= p1 - 1e-5), "extra hop lowered access prob"
