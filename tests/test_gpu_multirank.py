@@ -132,6 +132,9 @@ def test_dist_p2p_feature_build(tmp_path):
 # ---------------------------------------------------------------------------
 
 def _rccl_worker(rank, world, q, rdv_file):
+    # NOTE: RCCL (2.26) keeps NCCL's one-rank-per-device rule —
+    # ncclCommInitRank returns "invalid usage" for co-located ranks — so
+    # this worker only runs with world == min(2, ngpus).
     try:
         dev = rank % torch.cuda.device_count()
         torch.cuda.set_device(dev)
@@ -143,7 +146,7 @@ def _rccl_worker(rank, world, q, rdv_file):
         dist.broadcast_object_list(objs, src=0)
         comm = NcclComm(rank, world, objs[0], hosts=world, rank_per_host=1)
 
-        # paired send/recv inside one group (safe for co-located ranks)
+        # paired send/recv inside one group (works for a self-pair too)
         t_send = torch.full((64,), float(rank + 1), device=dev)
         t_recv = torch.zeros(64, device=dev)
         peer = (rank + 1) % world
@@ -160,18 +163,28 @@ def _rccl_worker(rank, world, q, rdv_file):
         expect = sum(r + 1 for r in range(world))
         assert torch.all(a == float(expect)), "allreduce sum"
 
-        # the two-phase exchange protocol over real RCCL send/recv
-        feat = torch.arange(100, dtype=torch.float32,
-                            device=dev).view(50, 2) * (rank + 1)
-        host2ids = [torch.tensor([1 + rank, 7, 11 + rank])
-                    for _ in range(world)]
-        feats = comm.exchange(host2ids, feat)
+        src = torch.full((world * 4,), float(rank), device=dev)
+        dst = torch.zeros(world * world * 4, device=dev)
+        comm.allgather(src, dst)
         torch.cuda.synchronize()
-        # the remote host served our ids from ITS feature tensor
-        expect = (torch.arange(100, dtype=torch.float32).view(50, 2)
-                  * (peer + 1))[host2ids[peer]]
-        got = feats[peer].cpu()
-        assert torch.equal(got, expect), f"exchange rows: {got} vs {expect}"
+        want = torch.arange(world, dtype=torch.float32) \
+            .repeat_interleave(world * 4)
+        assert torch.equal(dst.cpu(), want), "allgather layout"
+
+        if world > 1:
+            # the two-phase exchange protocol over real RCCL send/recv
+            feat = torch.arange(100, dtype=torch.float32,
+                                device=dev).view(50, 2) * (rank + 1)
+            host2ids = [torch.tensor([1 + rank, 7, 11 + rank])
+                        for _ in range(world)]
+            feats = comm.exchange(host2ids, feat)
+            torch.cuda.synchronize()
+            # the remote host served our ids from ITS feature tensor
+            expect = (torch.arange(100, dtype=torch.float32).view(50, 2)
+                      * (peer + 1))[host2ids[peer]]
+            got = feats[peer].cpu()
+            assert torch.equal(got, expect), \
+                f"exchange rows: {got} vs {expect}"
         dist.barrier()
         q.put((rank, "ok"))
         dist.destroy_process_group()
@@ -179,6 +192,10 @@ def _rccl_worker(rank, world, q, rdv_file):
         q.put((rank, f"{type(e).__name__}: {e}"))
 
 
-def test_rccl_two_ranks(tmp_path):
+def test_rccl_native_comm(tmp_path):
+    """2-rank native-RCCL flow on >=2 GPUs; on a 1-GPU box a 1-rank
+    self-communicator still covers the native init/send/recv/allreduce/
+    allgather path (one rank per device is an RCCL rule)."""
     os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
-    _run_workers(_rccl_worker, (str(tmp_path / "rdv"),))
+    world = min(WORLD, torch.cuda.device_count())
+    _run_workers(_rccl_worker, (str(tmp_path / "rdv"),), world=world)
