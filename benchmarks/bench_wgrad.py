@@ -4,12 +4,16 @@ tall-skinny shapes the bench's SAGE/GAT layers hit.
 
 C[M x N] = A^T @ B, A [K x M] = grad_out, B [K x N] = layer input.
 """
+import os
+import sys
 import time
 
 import torch
 
-import quiver  # noqa: F401
-from quiver import _ext
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import quiver  # noqa: F401,E402
+from quiver import _ext  # noqa: E402
 
 
 def time_fn(fn, iters=50, warmup=10):

@@ -1120,6 +1120,8 @@ torch::Tensor segment_mean_gather_backward(torch::Tensor grad_out,
 
 // Tall-skinny weight gradient: (A^T @ B, optional column-sums of A).
 // A [K×M] (grad_out), B [K×N] (layer input), K = frontier size.
+// Deterministic: split-K partials land in a workspace (torch pool
+// scratch) and are reduced in fixed order.
 std::tuple<torch::Tensor, torch::Tensor> wgrad(torch::Tensor a,
                                                torch::Tensor b,
                                                bool want_bias) {
@@ -1132,11 +1134,14 @@ std::tuple<torch::Tensor, torch::Tensor> wgrad(torch::Tensor a,
     b = b.contiguous();
     int64_t k = a.size(0);
     int m = (int)a.size(1), n = (int)b.size(1);
-    auto c = torch::zeros({m, n}, a.options());
-    auto bias = want_bias ? torch::zeros({m}, a.options()) : torch::Tensor();
+    auto c = torch::empty({m, n}, a.options());
+    auto bias = want_bias ? torch::empty({m}, a.options()) : torch::Tensor();
+    auto plan = qk::wgrad_plan(k, m, n);
+    auto ws = torch::empty({plan.ws_floats}, a.options());
     qk::launch_wgrad(current_stream(), a.data_ptr<float>(),
                      b.data_ptr<float>(), c.data_ptr<float>(),
-                     want_bias ? bias.data_ptr<float>() : nullptr, k, m, n);
+                     want_bias ? bias.data_ptr<float>() : nullptr, k, m, n,
+                     plan, ws.data_ptr<float>());
     return {c, bias};
 }
 

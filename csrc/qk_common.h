@@ -135,11 +135,19 @@ void launch_scatter(hipStream_t s, const GatherSpec& spec,
 
 // ---------- tall-skinny weight-grad GEMM (wgrad_kernels.hip) -------------
 
-// C[M×N] += A^T @ B with A [K×M], B [K×N] row-major, K huge; C (and
-// bias_grad, = column sums of A, if non-null) must be pre-zeroed.
-// Split-K with fp32 global atomics: reduction order nondeterministic.
+// C[M×N] = A^T @ B with A [K×M], B [K×N] row-major, K huge.  MFMA
+// split-K: per-chunk partials land in a caller-provided workspace and a
+// fixed-order reduce writes C — deterministic, no pre-zeroing needed.
+struct WgradPlan {
+    int tm, tn;         // C tile grid
+    int64_t nchunks;    // split-K factor
+    int64_t k_chunk;    // K rows per chunk
+    int64_t ws_floats;  // workspace floats: nchunks * padded C (+ bias col)
+};
+WgradPlan wgrad_plan(int64_t K, int M, int N);
 void launch_wgrad(hipStream_t s, const float* A, const float* B, float* C,
-                  float* bias_grad, int64_t K, int M, int N);
+                  float* bias_grad, int64_t K, int M, int N,
+                  const WgradPlan& plan, float* ws);
 
 // ---------- fused message-passing aggregation (segment_kernels.hip) ------
 

@@ -514,9 +514,9 @@ void test_wgrad() {
     auto* d_bias = dalloc<float>(M);
     h2d(d_a, a);
     h2d(d_b, b);
-    QK_CHECK_HIP(hipMemset(d_c, 0, sizeof(float) * M * N));
-    QK_CHECK_HIP(hipMemset(d_bias, 0, sizeof(float) * M));
-    qk::launch_wgrad(nullptr, d_a, d_b, d_c, d_bias, K, M, N);
+    auto plan = qk::wgrad_plan(K, M, N);
+    auto* d_ws = dalloc<float>((size_t)plan.ws_floats);
+    qk::launch_wgrad(nullptr, d_a, d_b, d_c, d_bias, K, M, N, plan, d_ws);
     QK_CHECK_HIP(hipDeviceSynchronize());
     auto c = d2h(d_c, (size_t)M * N);
     auto bias = d2h(d_bias, M);
@@ -535,6 +535,7 @@ void test_wgrad() {
             REQUIRE(std::abs(c[mm * N + nn] - acc) < tol);
         }
     hipFree(d_a); hipFree(d_b); hipFree(d_c); hipFree(d_bias);
+    hipFree(d_ws);
     printf("ok test_wgrad\n");
 }
 

@@ -57,3 +57,17 @@ def test_qlinear_autograd_matches_nn_linear():
     assert torch.allclose(ql.weight.grad, ref.weight.grad,
                           atol=1e-3, rtol=1e-3)
     assert torch.allclose(ql.bias.grad, ref.bias.grad, atol=1e-3, rtol=1e-3)
+
+
+def test_wgrad_deterministic():
+    """Split-K partials are reduced in fixed order (workspace + reduce
+    kernel), so repeated calls on identical inputs are bitwise equal —
+    unlike rocBLAS GSU."""
+    g = torch.Generator(device="cuda").manual_seed(3)
+    a = torch.randn(120_000, 256, device="cuda", generator=g)
+    b = torch.randn(120_000, 100, device="cuda", generator=g)
+    c0, bias0 = _ext.wgrad(a, b, True)
+    for _ in range(3):
+        c, bias = _ext.wgrad(a, b, True)
+        assert torch.equal(c, c0)
+        assert torch.equal(bias, bias0)
