@@ -28,6 +28,12 @@ def time_fn(fn, iters=50, warmup=10):
 
 
 def main():
+    import argparse
+    p = argparse.ArgumentParser()
+    p.add_argument("--shape", default=None,
+                   help="K,M,N: profile a single shape")
+    p.add_argument("--iters", type=int, default=50)
+    args = p.parse_args()
     shapes = [
         (90_000, 256, 100),   # products L1 wgrad (agg path)
         (90_000, 256, 256),
@@ -35,14 +41,16 @@ def main():
         (13_000, 256, 256),   # L2
         (33_000, 47, 256),    # last layer
     ]
+    if args.shape:
+        shapes = [tuple(int(x) for x in args.shape.split(","))]
     g = torch.Generator(device="cuda").manual_seed(0)
     print(f"{'K':>8} {'M':>4} {'N':>4} | {'mfma us':>9} {'torch us':>9} "
           f"{'speedup':>7} {'mfma TF':>8} {'rel err':>9}")
     for k, m, n in shapes:
         a = torch.randn(k, m, device="cuda", generator=g)
         b = torch.randn(k, n, device="cuda", generator=g)
-        t_q = time_fn(lambda: _ext.wgrad(a, b, True))
-        t_t = time_fn(lambda: (a.t() @ b, a.sum(0)))
+        t_q = time_fn(lambda: _ext.wgrad(a, b, True), iters=args.iters)
+        t_t = time_fn(lambda: (a.t() @ b, a.sum(0)), iters=args.iters)
         c, bias = _ext.wgrad(a, b, True)
         want = a.double().t() @ b.double()
         rel = float((c.double() - want).norm() / want.norm())
