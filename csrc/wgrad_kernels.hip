@@ -4,26 +4,32 @@
 //
 // rocBLAS/Tensile handles this shape with stream-K 32x32 macro-tiles ~60x
 // off the HBM3E roof; a VALU split-K ran it at ~14 TF.  This is a split-K
-// GEMM on the f32 matrix cores (v_mfma_f32_16x16x4_f32, exact f32
-// numerics, 157 TF chip peak), shaped by three observations:
+// GEMM on the f32 matrix cores (v_mfma_f32_32x32x2_f32, exact f32
+// numerics, 157 TF chip peak; measured 41-64 TF on the bench shapes =
+// 2.8-5.7x rocBLAS), shaped by four observations:
 //
-//  - The A^T shape is a gift on CDNA4: the MFMA A-fragment for a 16x16x4
-//    tile wants lane l to hold A_mfma[m = l%16][k = l/16], which for our
-//    K-major A is A[(k0 + l/16)*M + m0 + l%16] — 16 consecutive lanes
-//    read 16 consecutive floats.  Both operands stage through LDS
-//    (row-padded +4 floats: the 4 k-group lanes of a fragment read rows
-//    64 floats apart, which would 2-way-conflict unpadded) purely for
-//    cross-wave reuse; no transpose anywhere.  Staging is float4 and
-//    double-buffered through registers: the next K-slice's global loads
-//    issue before the MFMA loop of the current slice.
+//  - The A^T shape is a gift on CDNA4: the MFMA A-fragment wants lane l
+//    to hold A_mfma[col = l%32][k = l/32], which for our K-major A is a
+//    fully-coalesced read — both operands stage through LDS with no
+//    transpose anywhere, purely for cross-wave reuse.  The 32x32x2 form
+//    is used over 16x16x4 because its 32-lane fragment groups read ONE
+//    LDS row each -> inherently bank-conflict-free (PMC showed 2
+//    conflicts/MFMA on the 16x16x4 4-k-group layout), and it halves the
+//    instruction-issue load.  Staging is float4 and double-buffered both
+//    through registers (next K-slice's global loads issue before the
+//    MFMA loop) and through ping-pong LDS (one barrier per stage).
 //  - C is tiny (a few hundred KB) while the split-K grid is huge, so
 //    atomic partials would pile ~300 ops on every C word.  Instead each
 //    block stores its 64x64 partial to a per-chunk workspace slice
 //    (plain coalesced stores) and a fixed-order reduce kernel writes C —
 //    which also makes wgrad DETERMINISTIC, unlike Tensile's GSU path.
-//  - Block = 4 waves = a 64x64 C macro-tile; each wave owns a 2x2 grid
-//    of 16x16 MFMA tiles (4 independent f32x4 accumulators hide the
-//    40-cycle dependent-accumulator latency at the 32-cycle issue rate).
+//  - The grid is 1-D and XCD-aware: blocks that re-read the same A/B
+//    slice are mapped to the same XCD (hardware dispatches block i to
+//    XCD i%8), so the tn x A + tm x B read amplification is absorbed by
+//    that XCD's L2 instead of HBM (~9% measured).
+//  - Block = 4 waves = a 64x64 C macro-tile; each wave owns one 32x32
+//    MFMA tile (a single f32x16 accumulator chain reaches the 64-cycle
+//    issue rate per the ISA's dependent-latency table).
 //
 // The bias gradient (column sums of A) folds into the same pass: blocks
 // in the first N-tile column accumulate their LDS A-slices into an extra
@@ -84,16 +90,33 @@ __device__ __forceinline__ void store_slice(float (*S)[LDP], int tid,
     }
 }
 
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
 __global__ void __launch_bounds__(NTHREADS)
 wgrad_mfma_kernel(const float* __restrict__ A, const float* __restrict__ B,
                   float* __restrict__ ws, int64_t K, int M, int N,
-                  int64_t k_chunk, int do_bias) {
-    __shared__ float As[BK][LDP];
-    __shared__ float Bs[BK][LDP];
+                  int64_t k_chunk, int do_bias, int tm_, int tn_,
+                  int64_t nchunks_) {
+    // ping-pong LDS: one barrier per stage (write next buffer while the
+    // current one is being read)
+    __shared__ float As[2][BK][LDP];
+    __shared__ float Bs[2][BK][LDP];
 
-    const int m0 = blockIdx.x * BM;
-    const int n0 = blockIdx.y * BN;
-    const int64_t k_beg = (int64_t)blockIdx.z * k_chunk;
+    // XCD-aware tile mapping (1-D grid): the hardware dispatches block i
+    // to XCD i%8, so make each XCD own whole (mi, ni) groups of one K
+    // chunk — the tn blocks re-reading an A slice and the tm blocks
+    // re-reading a B slice then share that XCD's L2 instead of pulling
+    // from HBM again.  nchunks is padded to a multiple of 8 by the plan.
+    const int n_xcd = 8;
+    const int xcd = blockIdx.x % n_xcd;
+    const int slot = blockIdx.x / n_xcd;
+    const int tiles = tm_ * tn_;
+    const int rem = slot % tiles;
+    const int64_t z = xcd + (int64_t)n_xcd * (slot / tiles);
+    if (z >= nchunks_) return;  // padding block
+    const int m0 = (rem / tn_) * BM;
+    const int n0 = (rem % tn_) * BN;
+    const int64_t k_beg = z * k_chunk;
     const int64_t k_end = min(K, k_beg + k_chunk);
 
     const int tid = threadIdx.x;
@@ -101,14 +124,17 @@ wgrad_mfma_kernel(const float* __restrict__ A, const float* __restrict__ B,
     const int lane = tid % 64;
     const int wm = (wave % 2) * 32;  // wave's m offset in the block tile
     const int wn = (wave / 2) * 32;  // wave's n offset
-    const int fcol = lane % 16;      // fragment column (m or n)
-    const int fk = lane / 16;        // fragment k row (0..3)
+    // 32x32x2 fragment mapping: lane l holds op[col = l%32][k = l/32].
+    // Within a 32-lane bank-conflict group all lanes read ONE LDS row at
+    // consecutive addresses -> inherently conflict-free (the 16x16x4
+    // 4-k-group layout 2-way-conflicted: PMC showed 2 conflicts/MFMA).
+    const int fcol = lane % 32;
+    const int fk = lane / 32;
 
-    f32x4 acc00 = {0, 0, 0, 0}, acc01 = {0, 0, 0, 0};
-    f32x4 acc10 = {0, 0, 0, 0}, acc11 = {0, 0, 0, 0};
+    f32x16 acc = {};
 
     // bias fold: thread t accumulates column (t%BM) over its k-subrows
-    const bool bias_block = do_bias && (blockIdx.y == 0);
+    const bool bias_block = do_bias && (n0 == 0);
     const int bm = tid % BM;       // bias column
     const int bk0 = tid / BM;      // first k-subrow (stride NTHREADS/BM)
     float bsum = 0.f;
@@ -116,93 +142,102 @@ wgrad_mfma_kernel(const float* __restrict__ A, const float* __restrict__ B,
     float4 ra[F4_PER_SLICE], rb[F4_PER_SLICE];
     load_slice(A, k_beg, k_end, M, m0, tid, ra);
     load_slice(B, k_beg, k_end, N, n0, tid, rb);
+    store_slice(As[0], tid, ra);
+    store_slice(Bs[0], tid, rb);
+    __syncthreads();
 
+    int buf = 0;
     for (int64_t k0 = k_beg; k0 < k_end; k0 += BK) {
-        store_slice(As, tid, ra);
-        store_slice(Bs, tid, rb);
-        __syncthreads();
+        const bool more = k0 + BK < k_end;
         // issue the next slice's global loads before the MFMA loop
-        if (k0 + BK < k_end) {
+        if (more) {
             load_slice(A, k0 + BK, k_end, M, m0, tid, ra);
             load_slice(B, k0 + BK, k_end, N, n0, tid, rb);
         }
 #pragma unroll
-        for (int kk = 0; kk < BK; kk += 4) {
-            const float a0 = As[kk + fk][wm + fcol];
-            const float a1 = As[kk + fk][wm + 16 + fcol];
-            const float b0 = Bs[kk + fk][wn + fcol];
-            const float b1 = Bs[kk + fk][wn + 16 + fcol];
-            acc00 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc00,
-                                                         0, 0, 0);
-            acc01 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc01,
-                                                         0, 0, 0);
-            acc10 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc10,
-                                                         0, 0, 0);
-            acc11 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc11,
-                                                         0, 0, 0);
+        for (int kk = 0; kk < BK; kk += 2) {
+            const float a = As[buf][kk + fk][wm + fcol];
+            const float b = Bs[buf][kk + fk][wn + fcol];
+            acc = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc, 0, 0, 0);
         }
         if (bias_block) {
 #pragma unroll
             for (int kk = bk0; kk < BK; kk += NTHREADS / BM)
-                bsum += As[kk][bm];
+                bsum += As[buf][kk][bm];
+        }
+        if (more) {
+            store_slice(As[buf ^ 1], tid, ra);
+            store_slice(Bs[buf ^ 1], tid, rb);
         }
         __syncthreads();
+        buf ^= 1;
     }
 
     // workspace slice for this K-chunk: padded C tile grid + bias column.
     // Plain coalesced stores; a fixed-order reduce kernel folds chunks.
-    const int Np = (int)gridDim.y * BN;
+    // 32x32 D mapping: col = lane%32, row = (r&3) + 8*(r>>2) + 4*(lane/32).
+    const int Np = tn_ * BN;
     const int64_t row_pitch = Np + 1;  // +1: bias column at Np
-    float* slice = ws + (int64_t)blockIdx.z * ((int64_t)gridDim.x * BM) *
-                            row_pitch;
-    const int n_out0 = n0 + wn + fcol;
-    const int n_out1 = n0 + wn + 16 + fcol;
+    float* slice = ws + z * ((int64_t)tm_ * BM) * row_pitch;
+    const int n_out = n0 + wn + fcol;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-        const int mr0 = m0 + wm + fk * 4 + r;
-        const int mr1 = mr0 + 16;
-        slice[(int64_t)mr0 * row_pitch + n_out0] = acc00[r];
-        slice[(int64_t)mr0 * row_pitch + n_out1] = acc01[r];
-        slice[(int64_t)mr1 * row_pitch + n_out0] = acc10[r];
-        slice[(int64_t)mr1 * row_pitch + n_out1] = acc11[r];
+    for (int r = 0; r < 16; ++r) {
+        const int m_out = m0 + wm + (r & 3) + 8 * (r >> 2) + 4 * fk;
+        slice[(int64_t)m_out * row_pitch + n_out] = acc[r];
     }
     if (bias_block) {
         // 4 threads share a bias column: reduce through LDS (As is free)
-        (&As[0][0])[tid] = bsum;
+        (&As[0][0][0])[tid] = bsum;
         __syncthreads();
         if (tid < BM) {
             float s = 0.f;
 #pragma unroll
             for (int g = 0; g < NTHREADS / BM; ++g)
-                s += (&As[0][0])[tid + g * BM];
+                s += (&As[0][0][0])[tid + g * BM];
             slice[(int64_t)(m0 + tid) * row_pitch + Np] = s;
         }
     }
 }
 
 // C[m][n] = sum_z ws[z][m][n]; bias_grad[m] = sum_z ws[z][m][Np].
-// Fixed summation order -> deterministic result.
-__global__ void __launch_bounds__(256)
+// Fixed z-to-group assignment and fixed group order -> deterministic.
+// Parallelized over (output, z-group): a one-thread-per-output reduce
+// chains nchunks ~600ns strided loads and ran as long as the GEMM
+// itself (83 us measured); 16 z-groups cut the chain 16x and an LDS
+// combine folds the groups.
+constexpr int ROUT = 64;  // outputs per block
+constexpr int ZG = 16;    // parallel z groups
+
+__global__ void __launch_bounds__(ROUT * ZG)
 wgrad_reduce_kernel(const float* __restrict__ ws, float* __restrict__ C,
                     float* __restrict__ bias_grad, int M, int N, int Mp,
                     int Np, int64_t nchunks) {
+    __shared__ float part[ZG][ROUT];
     const int64_t row_pitch = Np + 1;
     const int64_t slice_sz = (int64_t)Mp * row_pitch;
     const int64_t total = (int64_t)M * N + (bias_grad ? M : 0);
-    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         i < total; i += (int64_t)gridDim.x * blockDim.x) {
-        int64_t off;
+    const int j = threadIdx.x % ROUT;
+    const int g = threadIdx.x / ROUT;
+    const int64_t i = (int64_t)blockIdx.x * ROUT + j;
+    float s = 0.f;
+    int64_t off = 0;
+    if (i < total) {
+        off = i < (int64_t)M * N
+                  ? (i / N) * row_pitch + (i % N)
+                  : (i - (int64_t)M * N) * row_pitch + Np;
+        for (int64_t z = g; z < nchunks; z += ZG)
+            s += ws[z * slice_sz + off];
+    }
+    part[g][j] = s;
+    __syncthreads();
+    if (g == 0 && i < total) {
+        float acc = 0.f;
+#pragma unroll
+        for (int gg = 0; gg < ZG; ++gg) acc += part[gg][j];
         if (i < (int64_t)M * N) {
-            off = (i / N) * row_pitch + (i % N);
+            C[i] = acc;
         } else {
-            off = (i - (int64_t)M * N) * row_pitch + Np;
-        }
-        float s = 0.f;
-        for (int64_t z = 0; z < nchunks; ++z) s += ws[z * slice_sz + off];
-        if (i < (int64_t)M * N) {
-            C[i] = s;
-        } else {
-            bias_grad[i - (int64_t)M * N] = s;
+            bias_grad[i - (int64_t)M * N] = acc;
         }
     }
 }
@@ -231,14 +266,16 @@ void launch_wgrad(hipStream_t s, const float* A, const float* B, float* C,
                   float* bias_grad, int64_t K, int M, int N,
                   const WgradPlan& plan, float* ws) {
     if (K == 0 || M == 0 || N == 0) return;
-    dim3 grid(plan.tm, plan.tn, (unsigned)plan.nchunks);
-    wgrad_mfma_kernel<<<grid, NTHREADS, 0, s>>>(A, B, ws, K, M, N,
-                                                plan.k_chunk,
-                                                bias_grad != nullptr);
+    // 1-D grid over XCD-padded chunks x tiles (see kernel mapping)
+    int64_t z_pad = (plan.nchunks + 7) / 8 * 8;
+    int64_t nblocks = z_pad * plan.tm * plan.tn;
+    wgrad_mfma_kernel<<<(unsigned)nblocks, NTHREADS, 0, s>>>(
+        A, B, ws, K, M, N, plan.k_chunk, bias_grad != nullptr, plan.tm,
+        plan.tn, plan.nchunks);
     QK_CHECK_HIP(hipGetLastError());
     int64_t total = (int64_t)M * N + (bias_grad ? M : 0);
-    int rblocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
-    wgrad_reduce_kernel<<<rblocks, 256, 0, s>>>(
+    int rblocks = (int)((total + ROUT - 1) / ROUT);
+    wgrad_reduce_kernel<<<rblocks, ROUT * ZG, 0, s>>>(
         ws, C, bias_grad, M, N, plan.tm * BM, plan.tn * BN, plan.nchunks);
     QK_CHECK_HIP(hipGetLastError());
 }
