@@ -255,7 +255,10 @@ class GpuSampler {
         for (void* p : registered_) (void)hipHostUnregister(p);
     }
 
-    void set_seed(uint64_t seed) { rng_counter_.store(seed); }
+    void set_seed(uint64_t seed) {
+        rng_counter_.store(seed);
+        write_rng(seed);
+    }
 
     std::tuple<torch::Tensor, torch::Tensor> sample_neighbor(
         int /*stream_id*/, torch::Tensor seeds, int k) {
@@ -272,11 +275,11 @@ class GpuSampler {
                                  counts.data_ptr<int64_t>());
         auto [prefix, total] = exclusive_scan_total(counts, stream);
         auto out = torch::empty({total}, opts);
-        uint64_t rs = rng_counter_.fetch_add(0x9e3779b97f4a7c15ULL);
+        qk::launch_rng_bump(stream, rng_dev(stream));
         qk::launch_sample(stream, indptr_.data_ptr<int64_t>(), indices_dptr_,
                           nullptr, seeds.data_ptr<int64_t>(), n, k,
                           prefix.data_ptr<int64_t>(), out.data_ptr<int64_t>(),
-                          nullptr, rs);
+                          nullptr, 0, rng_dev(stream));
         return {out, counts};
     }
 
@@ -362,6 +365,9 @@ class GpuSampler {
 
         auto sizes_dev = torch::empty({2 * H}, opts);
         int64_t* sd = sizes_dev.data_ptr<int64_t>();
+        // one seed advance per batch, as a device-side graph node (the
+        // chain stays hipGraph-capturable; per-hop streams get a salt)
+        qk::launch_rng_bump(stream, rng_dev(stream));
 
         torch::Tensor cur = seeds;
         const int64_t* n_dev = nullptr;     // exact frontier count (device)
@@ -384,12 +390,13 @@ class GpuSampler {
                                       prefix.data_ptr<int64_t>(), n_ub,
                                       sd + 2 * h);  // m_h
             auto out = torch::empty({m_ub}, opts);
-            uint64_t rs = rng_counter_.fetch_add(0x9e3779b97f4a7c15ULL);
+            const uint64_t salt = (uint64_t)(h + 1) * 0xD1B54A32D192ED03ULL;
             qk::launch_sample(stream, indptr_.data_ptr<int64_t>(),
                               indices_dptr_, nullptr,
                               cur.data_ptr<int64_t>(), n_ub, k,
                               prefix.data_ptr<int64_t>(),
-                              out.data_ptr<int64_t>(), nullptr, rs, n_dev);
+                              out.data_ptr<int64_t>(), nullptr, salt,
+                              rng_dev(stream), n_dev);
 
             // reindex [cur ++ out] -> frontier + local col ids
             const int64_t total_ub = n_ub + m_ub;
@@ -487,6 +494,24 @@ class GpuSampler {
         return register_host_chunked(p, bytes, registered_);
     }
 
+    // lazily-allocated device RNG word (int64 tensor reinterpreted)
+    uint64_t* rng_dev(hipStream_t /*stream*/) {
+        if (!rng_dev_.defined()) {
+            rng_dev_ = torch::empty(
+                {1}, torch::TensorOptions().dtype(torch::kInt64).device(
+                         torch::Device(torch::kCUDA, device_)));
+            write_rng(rng_counter_.load());
+        }
+        return (uint64_t*)rng_dev_.data_ptr();
+    }
+
+    void write_rng(uint64_t v) {
+        if (!rng_dev_.defined()) return;  // picked up at first rng_dev()
+        DeviceScope g(device_);
+        QK_CHECK_HIP(hipMemcpy(rng_dev_.data_ptr(), &v, sizeof(v),
+                               hipMemcpyHostToDevice));
+    }
+
     int device_;
     bool dma_;
     bool has_eid_ = false;
@@ -496,6 +521,7 @@ class GpuSampler {
     int64_t* eid_dptr_ = nullptr;
     std::vector<void*> registered_;
     std::atomic<uint64_t> rng_counter_{0x853c49e6748fea9bULL};
+    torch::Tensor rng_dev_;
 };
 
 // ---------------------------------------------------------------------------
@@ -838,11 +864,17 @@ class ShardTensor {
             QK_CHECK_HIP(hipStreamWaitEvent(cur.stream(), ev.second, 0));
             // caching-allocator hazard: out/indices are used on `side`.
             // (Tensor::record_stream wants a masquerading-as-CUDA stream;
-            // go to the HIP allocator directly.)
-            c10::hip::HIPCachingAllocator::recordStream(
-                out.storage().data_ptr(), side);
-            c10::hip::HIPCachingAllocator::recordStream(
-                indices.storage().data_ptr(), side);
+            // go to the HIP allocator directly.)  Under hipGraph capture
+            // the tensors live in the graph pool with static lifetime —
+            // skip the bookkeeping (recordStream during capture throws).
+            hipStreamCaptureStatus cs = hipStreamCaptureStatusNone;
+            (void)hipStreamIsCapturing(cur.stream(), &cs);
+            if (cs == hipStreamCaptureStatusNone) {
+                c10::hip::HIPCachingAllocator::recordStream(
+                    out.storage().data_ptr(), side);
+                c10::hip::HIPCachingAllocator::recordStream(
+                    indices.storage().data_ptr(), side);
+            }
         } else {
             qk::launch_gather(cur.stream(), spec,
                               indices.data_ptr<int64_t>(), n,

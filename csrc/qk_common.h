@@ -60,7 +60,12 @@ void launch_sample(hipStream_t s, const int64_t* indptr, const int64_t* indices,
                    const int64_t* eid_base, const int64_t* seeds, int64_t n,
                    int k, const int64_t* prefix, int64_t* out_nbrs,
                    int64_t* out_eids, uint64_t rng_seed,
+                   const uint64_t* rng_dev = nullptr,
                    const int64_t* n_dev = nullptr);
+
+// *rng += golden-ratio step; the per-batch seed advance as a graph node
+// (a captured chain must advance the seed ON DEVICE each replay).
+void launch_rng_bump(hipStream_t s, uint64_t* rng);
 
 // Access-probability propagation (one hop):
 // cur[v] = 1 - (1 - last[v]) * prod_{u in N(v)} (1 - last[u]*min(1, k/deg(u)))

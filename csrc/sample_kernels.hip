@@ -62,7 +62,11 @@ sample_kernel(const int64_t* __restrict__ indptr,
               const int64_t* __restrict__ seeds, int64_t n, int k,
               const int64_t* __restrict__ prefix,
               int64_t* __restrict__ out_nbrs, int64_t* __restrict__ out_eids,
-              uint64_t rng_seed, const int64_t* __restrict__ n_dev) {
+              uint64_t rng_seed, const uint64_t* __restrict__ rng_dev,
+              const int64_t* __restrict__ n_dev) {
+    // device-resident RNG state (hipGraph-capturable: a replayed graph
+    // must not freeze a host-read seed); rng_seed is the per-hop salt
+    if (rng_dev) rng_seed += *rng_dev;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     int* slots = reinterpret_cast<int*>(smem);  // ROWS_PER_BLOCK * k
 
@@ -209,7 +213,7 @@ void launch_sample(hipStream_t s, const int64_t* indptr, const int64_t* indices,
                    const int64_t* eid_base, const int64_t* seeds, int64_t n,
                    int k, const int64_t* prefix, int64_t* out_nbrs,
                    int64_t* out_eids, uint64_t rng_seed,
-                   const int64_t* n_dev) {
+                   const uint64_t* rng_dev, const int64_t* n_dev) {
     if (n == 0) return;
     if (k < 1) throw std::runtime_error("sample: fanout k must be >= 1");
     size_t lds = (size_t)ROWS_PER_BLOCK * k * sizeof(int);
@@ -220,12 +224,21 @@ void launch_sample(hipStream_t s, const int64_t* indptr, const int64_t* indices,
         sample_kernel<true><<<grid, BLOCK, lds, s>>>(indptr, indices, eid_base,
                                                      seeds, n, k, prefix,
                                                      out_nbrs, out_eids,
-                                                     rng_seed, n_dev);
+                                                     rng_seed, rng_dev, n_dev);
     else
         sample_kernel<false><<<grid, BLOCK, lds, s>>>(indptr, indices, eid_base,
                                                       seeds, n, k, prefix,
                                                       out_nbrs, nullptr,
-                                                      rng_seed, n_dev);
+                                                      rng_seed, rng_dev, n_dev);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+__global__ void rng_bump_kernel(uint64_t* rng) {
+    *rng += 0x9e3779b97f4a7c15ULL;
+}
+
+void launch_rng_bump(hipStream_t s, uint64_t* rng) {
+    rng_bump_kernel<<<1, 1, 0, s>>>(rng);
     QK_CHECK_HIP(hipGetLastError());
 }
 

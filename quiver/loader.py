@@ -17,11 +17,57 @@ Usage:
         out = model(x, adjs)
         ...
 """
+import os
 from collections import deque
 
 import torch
 
 __all__ = ["TrainingPrefetcher"]
+
+
+class _GraphedChain:
+    """`depth` alternating hipGraph captures of the sample->gather chain.
+
+    A captured graph replays the whole per-batch chain (~75 kernel
+    launches) as ONE hipGraphLaunch, eliminating per-launch host cost.
+    Requirements already guaranteed by the zero-sync chain design: no
+    host syncs anywhere (sizes land in pinned memory via an async D2H
+    node), upper-bound-sized static buffers, and a DEVICE-resident RNG
+    word advanced by a captured bump kernel (a host-read seed would be
+    frozen into the graph).
+
+    Each slot owns its own capture + static buffers: batch i and batch
+    i+1 are in flight simultaneously, so a single capture would overwrite
+    live outputs.  The caller replays slot i%depth for batch i and must
+    order each replay after the consumer of that slot's previous batch
+    (an event on the main stream).
+    """
+
+    def __init__(self, sampler, feature, seeds_proto, stream, depth):
+        self.batch_size = seeds_proto.numel()
+        self.slots = []
+        self.turn = 0
+        device = sampler.device
+        for _ in range(depth):
+            static_seeds = seeds_proto.to(device).clone()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.stream(stream):
+                with torch.cuda.graph(g, stream=stream):
+                    tok = sampler.sample_async(static_seeds)
+                    x_ub = None
+                    if feature is not None:
+                        raw, sizes_dev = tok[1], tok[2]
+                        n_dev = sizes_dev[2 * len(raw) - 1:2 * len(raw)]
+                        x_ub = feature.gather_raw(raw[-1][0], n_dev)
+            self.slots.append((static_seeds, g, tok, x_ub))
+
+    def run(self, seeds):
+        """Replay the next slot on the current stream."""
+        static_seeds, g, tok, x_ub = self.slots[self.turn]
+        self.turn = (self.turn + 1) % len(self.slots)
+        static_seeds.copy_(seeds, non_blocking=True)
+        g.replay()
+        return tok, x_ub
 
 
 class TrainingPrefetcher:
@@ -71,6 +117,12 @@ class TrainingPrefetcher:
         it = iter(self.seed_batches)
         pending = deque()
         chain_async = [self._can_chain_async()]
+        # hipGraph-replay the chain: default on for the async-chainable
+        # path with a single side stream; QUIVER_HIPGRAPH=0 opts out
+        want_graph = (chain_async[0] and self.num_streams == 1
+                      and os.environ.get("QUIVER_HIPGRAPH", "1") != "0")
+        graphed = [None]
+        produced = [0]
         rr = [0]
 
         def produce():
@@ -81,6 +133,40 @@ class TrainingPrefetcher:
             side = sides[rr[0] % len(sides)]
             rr[0] += 1
             with torch.cuda.stream(side):
+                if chain_async[0] and want_graph and graphed[0] is None \
+                        and produced[0] >= 1:
+                    # one eager batch has warmed the chain: capture
+                    try:
+                        graphed[0] = _GraphedChain(self.sampler,
+                                                   self.feature, seeds,
+                                                   side, self.depth)
+                    except Exception as e:  # noqa: BLE001
+                        import warnings
+                        warnings.warn("TrainingPrefetcher: hipGraph "
+                                      f"capture failed ({e}); running "
+                                      "uncaptured", RuntimeWarning)
+                        graphed[0] = False
+                if chain_async[0] and isinstance(graphed[0], _GraphedChain) \
+                        and seeds.numel() == graphed[0].batch_size:
+                    # order this slot's replay after the enqueued main-
+                    # stream work (incl. the training step of the batch
+                    # that previously used this slot's static buffers)
+                    ev_main = torch.cuda.Event()
+                    ev_main.record(cur)
+                    side.wait_event(ev_main)
+                    try:
+                        tok, x_ub = graphed[0].run(seeds)
+                        ev = torch.cuda.Event()
+                        ev.record(side)
+                        pending.append(("gtok", tok, x_ub, ev))
+                        produced[0] += 1
+                        return True
+                    except RuntimeError as e:
+                        import warnings
+                        warnings.warn("TrainingPrefetcher: hipGraph replay "
+                                      f"failed ({e}); running uncaptured",
+                                      RuntimeWarning)
+                        graphed[0] = False
                 if chain_async[0]:
                     try:
                         tok = self.sampler.sample_async(seeds)
@@ -92,6 +178,7 @@ class TrainingPrefetcher:
                         ev = torch.cuda.Event()
                         ev.record(side)
                         pending.append(("tok", tok, x_ub, ev))
+                        produced[0] += 1
                         return True
                     except RuntimeError as e:
                         # fall through to the sync path for the rest of the
@@ -117,27 +204,34 @@ class TrainingPrefetcher:
             kind, payload, x, ev = pending.popleft()
             # main stream waits for the side stream's work for THIS batch
             cur.wait_event(ev)
-            if kind == "tok":
+            if kind in ("tok", "gtok"):
                 # CPU-side wait for THIS batch only (event, not stream):
                 # makes the pinned sizes valid; later batches keep running
                 ev.synchronize()
                 n_id, bs, adjs = self.sampler.sample_finalize(payload)
-                # finalize's torch.stack reads the side-allocated raw
-                # buffers on the main stream
-                for f_ub, row_ub, col_ub in payload[1]:
-                    row_ub.record_stream(cur)
-                    col_ub.record_stream(cur)
                 if x is not None:
                     x = x[:n_id.size(0)]
             else:
                 n_id, bs, adjs = payload
-            # side-stream allocations must not be reused until main-stream
-            # work on them completes
-            n_id.record_stream(cur)
-            for adj in adjs:
-                adj.edge_index.record_stream(cur)
-            if x is not None:
-                x.record_stream(cur)
+            if kind == "gtok":
+                # graph-pool buffers have static lifetime: no allocator
+                # bookkeeping, and the next replay of this slot is ordered
+                # after this batch's training step by the producer's event
+                pass
+            else:
+                # side-stream allocations must not be reused until
+                # main-stream work on them completes
+                if kind == "tok":
+                    # finalize's torch.stack reads the side-allocated raw
+                    # buffers on the main stream
+                    for f_ub, row_ub, col_ub in payload[1]:
+                        row_ub.record_stream(cur)
+                        col_ub.record_stream(cur)
+                n_id.record_stream(cur)
+                for adj in adjs:
+                    adj.edge_index.record_stream(cur)
+                if x is not None:
+                    x.record_stream(cur)
             # consumer launches the training step for this batch inside the
             # yield; on re-entry we produce batch i+depth so the whole
             # sample+gather chain overlaps that compute
