@@ -48,17 +48,24 @@ class _GraphedChain:
         self.slots = []
         self.turn = 0
         device = sampler.device
+        H = len(sampler.sizes)
         for _ in range(depth):
             static_seeds = seeds_proto.to(device).clone()
+            # pinned buffer + its D2H copy stay OUTSIDE the capture:
+            # hipHostMalloc (and torch pin_memory allocation) is not a
+            # capturable operation
+            sizes_pin = torch.empty(2 * H, dtype=torch.int64,
+                                    pin_memory=True)
             g = torch.cuda.CUDAGraph()
             with torch.cuda.stream(stream):
                 with torch.cuda.graph(g, stream=stream):
-                    tok = sampler.sample_async(static_seeds)
+                    raw, sizes_dev = sampler.quiver.sample_hops_raw(
+                        static_seeds, sampler.sizes)
                     x_ub = None
                     if feature is not None:
-                        raw, sizes_dev = tok[1], tok[2]
-                        n_dev = sizes_dev[2 * len(raw) - 1:2 * len(raw)]
-                        x_ub = feature.gather_raw(raw[-1][0], n_dev)
+                        n_dev = sizes_dev[2 * H - 1:2 * H]
+                        x_ub = feature.gather_raw(raw[H - 1][0], n_dev)
+            tok = (static_seeds, raw, sizes_dev, sizes_pin)
             self.slots.append((static_seeds, g, tok, x_ub))
 
     def run(self, seeds):
@@ -67,6 +74,7 @@ class _GraphedChain:
         self.turn = (self.turn + 1) % len(self.slots)
         static_seeds.copy_(seeds, non_blocking=True)
         g.replay()
+        tok[3].copy_(tok[2], non_blocking=True)  # sizes D2H, post-replay
         return tok, x_ub
 
 
@@ -130,22 +138,27 @@ class TrainingPrefetcher:
                 seeds = next(it)
             except StopIteration:
                 return False
-            side = sides[rr[0] % len(sides)]
+            idx = rr[0] % len(sides)
             rr[0] += 1
+            if chain_async[0] and want_graph and graphed[0] is None \
+                    and produced[0] >= 1:
+                # one eager batch has warmed the chain: capture (outside
+                # the stream context — a failed capture can wedge the
+                # stream in capture state, in which case it is replaced)
+                try:
+                    graphed[0] = _GraphedChain(self.sampler, self.feature,
+                                               seeds, sides[idx],
+                                               self.depth)
+                except Exception as e:  # noqa: BLE001
+                    import warnings
+                    warnings.warn("TrainingPrefetcher: hipGraph capture "
+                                  f"failed ({e}); running uncaptured",
+                                  RuntimeWarning)
+                    graphed[0] = False
+                    sides[idx] = torch.cuda.Stream(self.device,
+                                                   priority=-1)
+            side = sides[idx]
             with torch.cuda.stream(side):
-                if chain_async[0] and want_graph and graphed[0] is None \
-                        and produced[0] >= 1:
-                    # one eager batch has warmed the chain: capture
-                    try:
-                        graphed[0] = _GraphedChain(self.sampler,
-                                                   self.feature, seeds,
-                                                   side, self.depth)
-                    except Exception as e:  # noqa: BLE001
-                        import warnings
-                        warnings.warn("TrainingPrefetcher: hipGraph "
-                                      f"capture failed ({e}); running "
-                                      "uncaptured", RuntimeWarning)
-                        graphed[0] = False
                 if chain_async[0] and isinstance(graphed[0], _GraphedChain) \
                         and seeds.numel() == graphed[0].batch_size:
                     # order this slot's replay after the enqueued main-
