@@ -49,7 +49,11 @@ class _GraphedChain:
         self.turn = 0
         device = sampler.device
         H = len(sampler.sizes)
-        for _ in range(depth):
+        # depth+1 slots: a replay then only needs the training step of
+        # the batch (depth+1) ago to have finished, which is one produce
+        # OLDER than the most recent main-stream event — so the replayed
+        # chain still overlaps the current batch's compute
+        for _ in range(depth + 1):
             static_seeds = seeds_proto.to(device).clone()
             # pinned buffer + its D2H copy stay OUTSIDE the capture:
             # hipHostMalloc (and torch pin_memory allocation) is not a
@@ -124,6 +128,7 @@ class TrainingPrefetcher:
                  for _ in range(self.num_streams)]
         it = iter(self.seed_batches)
         pending = deque()
+        main_evs = deque()
         chain_async = [self._can_chain_async()]
         # hipGraph-replay the chain: default on for the async-chainable
         # path with a single side stream; QUIVER_HIPGRAPH=0 opts out
@@ -161,12 +166,19 @@ class TrainingPrefetcher:
             with torch.cuda.stream(side):
                 if chain_async[0] and isinstance(graphed[0], _GraphedChain) \
                         and seeds.numel() == graphed[0].batch_size:
-                    # order this slot's replay after the enqueued main-
-                    # stream work (incl. the training step of the batch
-                    # that previously used this slot's static buffers)
+                    # order this slot's replay after the training step of
+                    # the batch that previously used this slot's static
+                    # buffers.  With depth+1 slots that step was enqueued
+                    # by the PREVIOUS produce call, so waiting on the
+                    # previous produce's main-stream event (not this
+                    # one's) keeps the replay overlapping the current
+                    # batch's compute.
                     ev_main = torch.cuda.Event()
                     ev_main.record(cur)
-                    side.wait_event(ev_main)
+                    main_evs.append(ev_main)
+                    if len(main_evs) > 2:
+                        main_evs.popleft()
+                    side.wait_event(main_evs[0])
                     try:
                         tok, x_ub = graphed[0].run(seeds)
                         ev = torch.cuda.Event()
