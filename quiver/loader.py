@@ -130,10 +130,14 @@ class TrainingPrefetcher:
         pending = deque()
         main_evs = deque()
         chain_async = [self._can_chain_async()]
-        # hipGraph-replay the chain: default on for the async-chainable
-        # path with a single side stream; QUIVER_HIPGRAPH=0 opts out
+        # hipGraph-replay the chain (QUIVER_HIPGRAPH=1 opt-in).  Measured
+        # on the products bench: capture works but replays ~7% SLOWER
+        # than stream launches (4.55 vs 4.24 ms/step, 40 steps) — the
+        # step is GPU-bound with CPU slack, so removing launch overhead
+        # buys nothing and graph scheduling costs a little.  Kept for
+        # launch-bound deployments (small batches / many hops).
         want_graph = (chain_async[0] and self.num_streams == 1
-                      and os.environ.get("QUIVER_HIPGRAPH", "1") != "0")
+                      and os.environ.get("QUIVER_HIPGRAPH", "0") == "1")
         graphed = [None]
         produced = [0]
         rr = [0]
