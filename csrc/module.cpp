@@ -340,8 +340,13 @@ class GpuSampler {
     // Removes ~12 stream syncs and ~40 python/torch ops per batch vs the
     // per-hop path — the training step was launch/GIL-bound, not
     // kernel-bound (profiles/: both streams <50% busy).
-    // Returns per hop: (frontier, row_idx, col_idx) with frontier[0:n_prev]
-    // == previous hop's frontier (seeds first), sizes already exact.
+    // Returns per hop: (frontier, edges) where edges is ONE [2, m] int64
+    // tensor with edges[0] = col_idx (src local ids) and edges[1] =
+    // row_idx (dst local ids) — the finished PyG edge_index with no
+    // torch.stack copy on the consumer's critical path (the stack of two
+    // ~1M-row hop buffers cost ~0.63 ms/step of main-stream time).
+    // frontier[0:n_prev] == previous hop's frontier (seeds first), sizes
+    // already exact.
     // Raw variant: returns UPPER-BOUND-sized per-hop tensors plus the
     // device-resident sizes vector [m_0, u_0, m_1, u_1, ...] WITHOUT any
     // host synchronization — callers chain the feature gather behind it
@@ -349,9 +354,8 @@ class GpuSampler {
     // read the sizes only when the whole chain has been consumed.
     // Frontier slack is zeroed so downstream indexing (feature_order
     // remap, gather) stays in-range.
-    std::tuple<
-        std::vector<std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>>,
-        torch::Tensor>
+    std::tuple<std::vector<std::tuple<torch::Tensor, torch::Tensor>>,
+               torch::Tensor>
     sample_hops_raw(torch::Tensor seeds, const std::vector<int>& ks) {
         DeviceScope g(device_);
         auto stream = current_stream();
@@ -372,7 +376,7 @@ class GpuSampler {
         torch::Tensor cur = seeds;
         const int64_t* n_dev = nullptr;     // exact frontier count (device)
         int64_t n_ub = seeds.numel();       // upper bound (host)
-        std::vector<torch::Tensor> frontiers, rows, cols;
+        std::vector<torch::Tensor> frontiers, hop_edges;
         std::vector<int64_t> n_ubs;
 
         for (int h = 0; h < H; ++h) {
@@ -433,45 +437,44 @@ class GpuSampler {
                 n_ub, out.data_ptr<int64_t>(), m_ub,
                 scanned.data_ptr<int64_t>(), flags.data_ptr<int64_t>(),
                 frontier.data_ptr<int64_t>(), n_dev, sd + 2 * h);
-            auto col_idx = torch::empty({m_ub}, opts);
+            // edges[0] = col (src local), edges[1] = row (dst local):
+            // the finished edge_index, written in place by the kernels
+            auto edges = torch::empty({2, m_ub}, opts);
+            int64_t* colp = edges.data_ptr<int64_t>();
+            int64_t* rowp = colp + m_ub;
             qk::launch_lookup_local(stream, keys.data_ptr<int64_t>(),
                                     local.data_ptr<int32_t>(), capacity,
                                     out.data_ptr<int64_t>(), m_ub,
-                                    col_idx.data_ptr<int64_t>(), sd + 2 * h);
-            auto row_idx = torch::empty({m_ub}, opts);
+                                    colp, sd + 2 * h);
             qk::launch_expand_rows(stream, prefix.data_ptr<int64_t>(),
                                    counts.data_ptr<int64_t>(), n_ub,
-                                   row_idx.data_ptr<int64_t>(), n_dev);
+                                   rowp, n_dev);
 
             frontiers.push_back(frontier);
-            rows.push_back(row_idx);
-            cols.push_back(col_idx);
+            hop_edges.push_back(edges);
             n_ubs.push_back(n_ub);
             cur = frontier;
             n_dev = sd + 2 * h + 1;
             n_ub = total_ub;
         }
 
-        std::vector<std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>>
-            raw;
+        std::vector<std::tuple<torch::Tensor, torch::Tensor>> raw;
         for (int h = 0; h < H; ++h)
-            raw.emplace_back(frontiers[h], rows[h], cols[h]);
+            raw.emplace_back(frontiers[h], hop_edges[h]);
         return {raw, sizes_dev};
     }
 
-    std::vector<std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>>
+    std::vector<std::tuple<torch::Tensor, torch::Tensor>>
     sample_hops(torch::Tensor seeds, const std::vector<int>& ks) {
         auto [raw, sizes_dev] = sample_hops_raw(seeds, ks);
         // the ONE sync of the whole batch
         auto sizes_host = sizes_dev.cpu();
         const int64_t* sh = sizes_host.data_ptr<int64_t>();
-        std::vector<std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>>
-            res;
+        std::vector<std::tuple<torch::Tensor, torch::Tensor>> res;
         for (size_t h = 0; h < raw.size(); ++h) {
             int64_t m = sh[2 * h], u = sh[2 * h + 1];
             res.emplace_back(std::get<0>(raw[h]).narrow(0, 0, u),
-                             std::get<1>(raw[h]).narrow(0, 0, m),
-                             std::get<2>(raw[h]).narrow(0, 0, m));
+                             std::get<1>(raw[h]).narrow(1, 0, m));
         }
         return res;
     }

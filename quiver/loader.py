@@ -251,11 +251,10 @@ class TrainingPrefetcher:
                 # side-stream allocations must not be reused until
                 # main-stream work on them completes
                 if kind == "tok":
-                    # finalize's torch.stack reads the side-allocated raw
-                    # buffers on the main stream
-                    for f_ub, row_ub, col_ub in payload[1]:
-                        row_ub.record_stream(cur)
-                        col_ub.record_stream(cur)
+                    # the edge_index views alias the side-allocated
+                    # upper-bound buffers read on the main stream
+                    for f_ub, e_ub in payload[1]:
+                        e_ub.record_stream(cur)
                 n_id.record_stream(cur)
                 for adj in adjs:
                     adj.edge_index.record_stream(cur)

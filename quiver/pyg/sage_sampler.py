@@ -118,8 +118,10 @@ class GraphSageSampler:
             with trace_scope("sampler.sample_hops"):
                 hops = self.quiver.sample_hops(nodes, self.sizes)
             prev_n = nodes.size(0)
-            for frontier, row_idx, col_idx in hops:
-                edge_index = torch.stack([col_idx, row_idx], dim=0)
+            for frontier, edge_index in hops:
+                # edge_index is emitted by the native chain as one [2, m]
+                # tensor (edges[0]=src local, edges[1]=dst local): no
+                # stack copy here
                 adj_size = torch.LongTensor([frontier.size(0), prev_n])
                 adjs.append(Adj(edge_index, torch.tensor([]), adj_size))
                 prev_n = frontier.size(0)
@@ -182,11 +184,12 @@ class GraphSageSampler:
         adjs = []
         prev = nodes.size(0)
         frontier = nodes
-        for h, (f_ub, row_ub, col_ub) in enumerate(raw):
+        for h, (f_ub, e_ub) in enumerate(raw):
             m = int(sizes_pin[2 * h])
             u = int(sizes_pin[2 * h + 1])
             frontier = f_ub[:u]
-            edge_index = torch.stack([col_ub[:m], row_ub[:m]], dim=0)
+            # zero-copy: a narrowed view of the upper-bound edge buffer
+            edge_index = e_ub.narrow(1, 0, m)
             adjs.append(Adj(edge_index, torch.tensor([]),
                             torch.LongTensor([u, prev])))
             prev = u
