@@ -103,6 +103,7 @@ class RequestBatcher(object):
     def fixed_despatch(self, idx):
         _enable_stack_dump()
         _pin(self.cpu_range, idx)
+        torch.set_num_threads(1)  # routing is numpy-light; don't spawn pools
         stream_queue = self.stream_queue_list[idx]
         if self.sample_mode == "CPU":
             batched_queue = self.cpu_batched_queue_list[idx % self.device_num]
@@ -113,11 +114,14 @@ class RequestBatcher(object):
             if isinstance(item, _Stop):
                 batched_queue.put(item)
                 break
-            batched_queue.put(item)
+            # arrival timestamp rides with the request so downstream p99
+            # includes queue wait (honest end-to-end latency)
+            batched_queue.put((item, time.perf_counter()))
 
     def auto_despatch(self, idx):
         _enable_stack_dump()
         _pin(self.cpu_range, idx)
+        torch.set_num_threads(1)
         stream_queue = self.stream_queue_list[idx]
         gpu_q = self.gpu_batched_queue_list[idx % self.device_num]
         cpu_q = self.cpu_batched_queue_list[idx % self.device_num]
@@ -129,11 +133,12 @@ class RequestBatcher(object):
                 cpu_q.put(item)
                 break
             if self.request_mode == "Preparation":
-                gpu_q.put(item)
-                cpu_q.put(item)
+                gpu_q.put((item, time.perf_counter()))
+                cpu_q.put((item, time.perf_counter()))
                 continue
             predicted_work = np.take(neighbour_num, np.asarray(item)).sum()
-            (gpu_q if predicted_work > self.threshold else cpu_q).put(item)
+            (gpu_q if predicted_work > self.threshold else cpu_q).put(
+                (item, time.perf_counter()))
 
     def batched_request_queue_list(self):
         return [self.cpu_batched_queue_list, self.gpu_batched_queue_list]
@@ -175,6 +180,10 @@ class HybridSampler(object):
                                 csr_topo):
         _enable_stack_dump()
         _pin(self.cpu_range, rank)
+        # single-threaded: a 64-seed request samples in ~3 ms on one core,
+        # while worker_count x default-OMP-pool oversubscription was the
+        # round-1 Auto-mode ~100 ms p99 (8 workers x 8+ threads thrashing)
+        torch.set_num_threads(1)
         cpu_sampler = GraphSageSampler(csr_topo, sizes, device="cpu",
                                        mode="CPU")
         task_queue = sample_task_queue_list[rank % device_num]
@@ -186,8 +195,10 @@ class HybridSampler(object):
                 task_queue.put(item)  # let sibling workers see it too
                 result_queue.put(item)
                 break
-            res = cpu_sampler.sample(item)
-            result_queue.put((res, time.perf_counter() - start))
+            ids, t_arrival = item if isinstance(item, tuple) else (item,
+                                                                   start)
+            res = cpu_sampler.sample(ids)
+            result_queue.put((res, time.perf_counter() - start, t_arrival))
 
     def sampled_request_queue_list(self):
         return [self.cpu_sampled_queue_list, self.gpu_batched_queue_list]
@@ -326,7 +337,8 @@ class InferenceServer(object):
                     q.put(item)
                     output_queue.put(item)
                     break
-                sample_task = torch.as_tensor(item)
+                ids = item[0] if isinstance(item, tuple) else item
+                sample_task = torch.as_tensor(ids)
                 n_id, batch_size, adjs = sampler.sample(sample_task)
                 adjs = [adj.to(device) for adj in adjs]
                 x_input = _feature_rows(feature, n_id, device)
@@ -345,7 +357,7 @@ class InferenceServer(object):
             adjs = [adj.to(device) for adj in adjs]
             x_input = _feature_rows(feature, n_id, device)
             out = model(x_input, adjs)
-            output_queue.put(out.cpu())
+            output_queue.put(out.cpu())  # item[2] (arrival) used by Debug
 
         with torch.no_grad():
             while True:
@@ -418,8 +430,10 @@ class InferenceServer_Debug(InferenceServer):
                                          rank, self.result_path, self.exp_id)
                     output_queue.put(stats if stats is not None else item)
                     break
-                start_time = time.perf_counter()
-                sample_task = torch.as_tensor(item)
+                pickup = time.perf_counter()
+                ids, start_time = (item if isinstance(item, tuple)
+                                   else (item, pickup))
+                sample_task = torch.as_tensor(ids)
                 n_id, batch_size, adjs = sampler.sample(sample_task)
                 sample_time = time.perf_counter()
                 adjs = [adj.to(device) for adj in adjs]
@@ -441,15 +455,16 @@ class InferenceServer_Debug(InferenceServer):
         result = []
 
         def infer(item):
-            start_time = time.perf_counter()
+            pickup = time.perf_counter()
             n_id, batch_size, adjs = item[0]
+            start_time = item[2] if len(item) > 2 else pickup
             adjs = [adj.to(device) for adj in adjs]
             x_input = _feature_rows(feature, n_id, device)
             out = model(x_input, adjs)
             if device != "cpu":
                 torch.cuda.synchronize()
             end_time = time.perf_counter()
-            result.append([start_time, start_time, end_time, batch_size,
+            result.append([start_time, pickup, end_time, batch_size,
                            n_id.shape[0]])
 
         with torch.no_grad():

@@ -74,8 +74,8 @@ def test_auto_routing_threshold(tmp_path):
     cpu_qs, gpu_qs = batcher.batched_request_queue_list()
     stream_queues[0].put(np.array([0, 1]))       # work 2000 -> GPU
     stream_queues[0].put(np.array([50, 51]))     # work 2 -> CPU
-    heavy = gpu_qs[0].get(timeout=30)
-    light = cpu_qs[0].get(timeout=30)
+    heavy, _t = gpu_qs[0].get(timeout=30)
+    light, _t2 = cpu_qs[0].get(timeout=30)
     assert list(heavy) == [0, 1]
     assert list(light) == [50, 51]
     batcher.stop()
@@ -105,9 +105,10 @@ def test_auto_despatch_routing(tmp_path):
 
     cpu_q = batcher.cpu_batched_queue_list[0]
     gpu_q = batcher.gpu_batched_queue_list[0]
-    got_cpu = cpu_q.get(timeout=10)
-    got_gpu = gpu_q.get(timeout=10)
+    got_cpu, t_cpu = cpu_q.get(timeout=10)
+    got_gpu, t_gpu = gpu_q.get(timeout=10)
     assert list(got_cpu) == [1, 2, 3]
     assert list(got_gpu) == [60, 61]
+    assert t_cpu > 0 and t_gpu > 0  # arrival stamps ride with requests
     assert isinstance(gpu_q.get(timeout=10), _Stop)
     assert isinstance(cpu_q.get(timeout=10), _Stop)
