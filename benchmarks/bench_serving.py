@@ -84,7 +84,7 @@ def main():
                    help="run everything on CPU (smoke test)")
     p.add_argument("--devices", type=int, nargs="*", default=None,
                    help="GPU ids for inference workers (default: all)")
-    p.add_argument("--procs-per-device", type=int, default=1)
+    p.add_argument("--procs-per-device", type=int, default=2)
     p.add_argument("--cpu-workers-per-device", type=int, default=8)
     p.add_argument("--nodes", type=int, default=N_NODES)
     p.add_argument("--edges", type=int, default=N_EDGES)
@@ -176,6 +176,8 @@ def main():
           f"seeds, mode={args.mode}, devices={device_list}", flush=True)
 
     server.start(join=False)
+    ready = server.wait_ready(timeout=180)
+    print(f"# {ready}/{server.num_proc} inference workers warm", flush=True)
     clients = []
     for i, sq in enumerate(stream_queues):
         c = mp.Process(target=client_loop,

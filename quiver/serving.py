@@ -271,6 +271,24 @@ class InferenceServer(object):
         self.num_proc = len(self.device_list) * self.proc_num_per_device
         self.output_queue_list = [_ctx.Queue()
                                   for _ in range(self.num_proc)]
+        # workers announce themselves here once model+sampler are warm;
+        # callers can gate offered load on wait_ready() so cold-start
+        # (model load, first CUDA context) doesn't become a request backlog
+        self.ready_queue = _ctx.Queue()
+
+    def wait_ready(self, timeout=180.0):
+        """Block until every inference worker finished warm-up."""
+        import queue as _queue
+        deadline = time.perf_counter() + timeout
+        seen = 0
+        while seen < self.num_proc:
+            try:
+                self.ready_queue.get(
+                    timeout=max(0.1, deadline - time.perf_counter()))
+                seen += 1
+            except _queue.Empty:
+                break
+        return seen
 
     def start(self, join=True):
         self.spawn_ctx = mp.spawn(self.run,
@@ -330,6 +348,7 @@ class InferenceServer(object):
                                            else "cpu"), mode=mode)
         model = _load_model(model_path, device)
         model.eval()
+        self._warmup_and_signal(sampler, feature, model, device)
         with torch.no_grad():
             while True:
                 item = q.get()
@@ -345,6 +364,25 @@ class InferenceServer(object):
                 out = model(x_input, adjs)
                 output_queue.put(out.cpu())
 
+    def _warmup_and_signal(self, sampler, feature, model, device):
+        """First-touch init (CUDA context, sampler engine, feature IPC)
+        before announcing readiness."""
+        try:
+            with torch.no_grad():
+                if sampler is not None:
+                    n_id, _, adjs = sampler.sample(torch.tensor([0, 1]))
+                    adjs = [adj.to(device) for adj in adjs]
+                    model(_feature_rows(feature, n_id, device), adjs)
+                else:
+                    _feature_rows(feature,
+                                  torch.tensor([0], dtype=torch.long),
+                                  device)
+                if device != "cpu":
+                    torch.cuda.synchronize()
+        except Exception:  # noqa: BLE001 - warm-up is best-effort
+            pass
+        self.ready_queue.put(os.getpid())
+
     def cpu_sampler_inference_loop(self, rank, device_list, feature,
                                    cpu_sampled_queue_list, model_path,
                                    output_queue):
@@ -352,6 +390,7 @@ class InferenceServer(object):
         q = cpu_sampled_queue_list[rank % len(device_list)]
         model = _load_model(model_path, device)
         model.eval()
+        self._warmup_and_signal(None, feature, model, device)
         def infer(item):
             n_id, batch_size, adjs = item[0]
             adjs = [adj.to(device) for adj in adjs]
@@ -420,6 +459,7 @@ class InferenceServer_Debug(InferenceServer):
                                            else "cpu"), mode=mode)
         model = _load_model(model_path, device)
         model.eval()
+        self._warmup_and_signal(sampler, feature, model, device)
         result = []
         with torch.no_grad():
             while True:
@@ -452,6 +492,7 @@ class InferenceServer_Debug(InferenceServer):
         q = cpu_sampled_queue_list[rank % len(device_list)]
         model = _load_model(model_path, device)
         model.eval()
+        self._warmup_and_signal(None, feature, model, device)
         result = []
 
         def infer(item):
