@@ -1123,12 +1123,22 @@ class ShardTensor {
 // ---------------------------------------------------------------------------
 torch::Tensor segment_mean_gather(torch::Tensor x, torch::Tensor src,
                                   torch::Tensor dst_ptr) {
-    TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kFloat32 && x.dim() == 2);
+    TORCH_CHECK(x.is_cuda() && x.dim() == 2 &&
+                (x.dtype() == torch::kFloat32 ||
+                 x.dtype() == torch::kBFloat16));
     x = x.contiguous();
     src = src.contiguous();
     dst_ptr = dst_ptr.contiguous();
     int64_t n_dst = dst_ptr.numel() - 1;
     auto out = torch::empty({n_dst, x.size(1)}, x.options());
+    if (x.dtype() == torch::kBFloat16) {
+        TORCH_CHECK(x.size(1) % 2 == 0,
+                    "bf16 segment mean needs an even feature dim");
+        qk::launch_segment_mean_fwd_bf16(
+            current_stream(), x.data_ptr(), src.data_ptr<int64_t>(),
+            dst_ptr.data_ptr<int64_t>(), n_dst, x.size(1), out.data_ptr());
+        return out;
+    }
     qk::launch_segment_mean_fwd(current_stream(), x.data_ptr<float>(),
                                 src.data_ptr<int64_t>(),
                                 dst_ptr.data_ptr<int64_t>(), n_dst, x.size(1),
@@ -1145,6 +1155,15 @@ torch::Tensor segment_mean_gather_backward(torch::Tensor grad_out,
     dst_ptr = dst_ptr.contiguous();
     int64_t n_dst = dst_ptr.numel() - 1;
     auto grad_x = torch::zeros({n_src, grad_out.size(1)}, grad_out.options());
+    if (grad_out.dtype() == torch::kBFloat16) {
+        TORCH_CHECK(grad_out.size(1) % 2 == 0,
+                    "bf16 segment mean needs an even feature dim");
+        qk::launch_segment_mean_bwd_bf16(
+            current_stream(), grad_out.data_ptr(), src.data_ptr<int64_t>(),
+            dst_ptr.data_ptr<int64_t>(), n_dst, grad_out.size(1),
+            grad_x.data_ptr());
+        return grad_x;
+    }
     qk::launch_segment_mean_bwd(current_stream(),
                                 grad_out.data_ptr<float>(),
                                 src.data_ptr<int64_t>(),

@@ -160,6 +160,13 @@ void launch_wgrad(hipStream_t s, const float* A, const float* B, float* C,
 void launch_segment_mean_fwd(hipStream_t s, const float* x,
                              const int64_t* src, const int64_t* dst_ptr,
                              int64_t n_dst, int64_t dim, float* out);
+// bf16 variants: bf16 rows, fp32 accumulation, packed-bf16 atomics (bwd)
+void launch_segment_mean_fwd_bf16(hipStream_t s, const void* x,
+                                  const int64_t* src, const int64_t* dst_ptr,
+                                  int64_t n_dst, int64_t dim, void* out);
+void launch_segment_mean_bwd_bf16(hipStream_t s, const void* grad_out,
+                                  const int64_t* src, const int64_t* dst_ptr,
+                                  int64_t n_dst, int64_t dim, void* grad_x);
 // grad_x[src[e]] += grad_out[d] / deg(d)   (grad_x pre-zeroed)
 void launch_segment_mean_bwd(hipStream_t s, const float* grad_out,
                              const int64_t* src, const int64_t* dst_ptr,

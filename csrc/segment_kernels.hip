@@ -10,6 +10,8 @@
 // Backward: grad_x[src[e]] += grad_out[d] / deg(d)  (device-scope f32
 // atomics; layer-1 input features skip it entirely since they carry no
 // grad).
+#include <hip/hip_bf16.h>
+
 #include "qk_common.h"
 
 namespace qk {
@@ -76,6 +78,97 @@ segment_mean_bwd_kernel(const float* __restrict__ grad_out,
             const float* orow = grad_out + d * dim;
             for (int64_t c = lane; c < dim; c += SUB)
                 atomicAdd(&grow[c], orow[c] * inv);
+        }
+    }
+}
+
+// ---- bf16 variants of the segment mean (fp32 accumulate) ----------------
+// Feature rows load as 16-byte bf16x8 chunks; accumulation stays fp32 and
+// only the store rounds.  Backward uses the packed-bf16 global atomic
+// (flat_atomic_fadd_v2bf16) via HIP's unsafeAtomicAdd — fine here: the
+// gradient buffer lives in ordinary HBM (no system-scope requirement).
+
+constexpr int VPL_BF = 8;  // channels per lane chunk (4x bf16x2 loads:
+                           // 4B-aligned for any even dim; a 16B vector
+                           // would need dim % 8 == 0)
+
+__global__ void __launch_bounds__(BLOCK)
+segment_mean_fwd_bf16_kernel(const __hip_bfloat16* __restrict__ x,
+                             const int64_t* __restrict__ src,
+                             const int64_t* __restrict__ dst_ptr,
+                             int64_t n_dst, int64_t dim,
+                             __hip_bfloat16* __restrict__ out) {
+    const int sub_id = threadIdx.x / SUB;
+    const int lane = threadIdx.x % SUB;
+    int64_t d = (int64_t)blockIdx.x * ROWS_PER_BLOCK + sub_id;
+    const int64_t stride = (int64_t)gridDim.x * ROWS_PER_BLOCK;
+    for (; d < n_dst; d += stride) {
+        const int64_t beg = dst_ptr[d], end = dst_ptr[d + 1];
+        const float inv = (end > beg) ? 1.0f / (float)(end - beg) : 0.0f;
+        for (int64_t c = (int64_t)lane * VPL_BF; c < dim; c += SUB * VPL_BF) {
+            float acc[VPL_BF] = {};
+            const int w = (int)min((int64_t)VPL_BF, dim - c);
+            for (int64_t e = beg; e < end; ++e) {
+                const __hip_bfloat16* row = x + src[e] * dim + c;
+                if (w == VPL_BF) {
+#pragma unroll
+                    for (int i = 0; i < 4; ++i) {
+                        const float2 f = __bfloat1622float2(
+                            *reinterpret_cast<const __hip_bfloat162*>(
+                                row + 2 * i));
+                        acc[2 * i] += f.x;
+                        acc[2 * i + 1] += f.y;
+                    }
+                } else {
+                    for (int q = 0; q < w; ++q) acc[q] += (float)row[q];
+                }
+            }
+            __hip_bfloat16* orow = out + d * dim + c;
+            if (w == VPL_BF) {
+#pragma unroll
+                for (int i = 0; i < 4; ++i)
+                    *reinterpret_cast<__hip_bfloat162*>(orow + 2 * i) =
+                        __float22bfloat162_rn(float2{acc[2 * i] * inv,
+                                                     acc[2 * i + 1] * inv});
+            } else {
+                for (int q = 0; q < w; ++q)
+                    orow[q] = (__hip_bfloat16)(acc[q] * inv);
+            }
+        }
+    }
+}
+
+__global__ void __launch_bounds__(BLOCK)
+segment_mean_bwd_bf16_kernel(const __hip_bfloat16* __restrict__ grad_out,
+                             const int64_t* __restrict__ src,
+                             const int64_t* __restrict__ dst_ptr,
+                             int64_t n_dst, int64_t dim,
+                             __hip_bfloat16* __restrict__ grad_x) {
+    const int sub_id = threadIdx.x / SUB;
+    const int lane = threadIdx.x % SUB;
+    int64_t d = (int64_t)blockIdx.x * ROWS_PER_BLOCK + sub_id;
+    const int64_t stride = (int64_t)gridDim.x * ROWS_PER_BLOCK;
+    for (; d < n_dst; d += stride) {
+        const int64_t beg = dst_ptr[d], end = dst_ptr[d + 1];
+        if (end <= beg) continue;
+        const float inv = 1.0f / (float)(end - beg);
+        for (int64_t e = beg; e < end; ++e) {
+            __hip_bfloat16* grow = grad_x + src[e] * dim;
+            const __hip_bfloat16* orow = grad_out + d * dim;
+            // packed atomic: lanes own even channel pairs
+            for (int64_t c = 2 * lane; c + 1 < dim; c += 2 * SUB) {
+                const float2 g = __bfloat1622float2(
+                    *reinterpret_cast<const __hip_bfloat162*>(&orow[c]));
+                const __hip_bfloat162 add = __float22bfloat162_rn(
+                    float2{g.x * inv, g.y * inv});
+                unsafeAtomicAdd(reinterpret_cast<__hip_bfloat162*>(&grow[c]),
+                                add);
+            }
+            if (lane == 0 && (dim & 1)) {
+                // odd tail channel
+                const float g = (float)orow[dim - 1] * inv;
+                unsafeAtomicAdd(&grow[dim - 1], (__hip_bfloat16)g);
+            }
         }
     }
 }
@@ -323,6 +416,28 @@ void launch_segment_mean_bwd(hipStream_t s, const float* grad_out,
     if (n_dst == 0) return;
     segment_mean_bwd_kernel<<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0, s>>>(
         grad_out, src, dst_ptr, n_dst, dim, grad_x);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+void launch_segment_mean_fwd_bf16(hipStream_t s, const void* x,
+                                  const int64_t* src, const int64_t* dst_ptr,
+                                  int64_t n_dst, int64_t dim, void* out) {
+    if (n_dst == 0) return;
+    segment_mean_fwd_bf16_kernel<<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0,
+                                   s>>>((const __hip_bfloat16*)x, src,
+                                        dst_ptr, n_dst, dim,
+                                        (__hip_bfloat16*)out);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+void launch_segment_mean_bwd_bf16(hipStream_t s, const void* grad_out,
+                                  const int64_t* src, const int64_t* dst_ptr,
+                                  int64_t n_dst, int64_t dim, void* grad_x) {
+    if (n_dst == 0) return;
+    segment_mean_bwd_bf16_kernel<<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0,
+                                   s>>>((const __hip_bfloat16*)grad_out, src,
+                                        dst_ptr, n_dst, dim,
+                                        (__hip_bfloat16*)grad_x);
     QK_CHECK_HIP(hipGetLastError());
 }
 

@@ -184,8 +184,11 @@ def _mean_aggregate(x_src, src, dst, n_dst, sorted_dst=False):
     """Mean of x_src[src] grouped by dst.  Uses the fused HIP kernel when
     the caller guarantees dst is sorted ascending (our sampler's layout)
     on fp32 GPU tensors."""
-    if (sorted_dst and x_src.is_cuda and x_src.dtype == torch.float32
-            and dst.numel() > 0):
+    fused_ok = (sorted_dst and x_src.is_cuda and dst.numel() > 0
+                and (x_src.dtype == torch.float32
+                     or (x_src.dtype == torch.bfloat16
+                         and x_src.size(1) % 2 == 0)))
+    if fused_ok:
         dst_ptr = _dst_ptr_from_sorted(dst, n_dst)
         return _SegmentMeanAgg.apply(x_src, src, dst_ptr)
     agg = torch.zeros((n_dst, x_src.size(1)), dtype=x_src.dtype,

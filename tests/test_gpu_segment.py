@@ -163,3 +163,61 @@ def test_segment_softmax_matches_torch():
     want.backward(go)
     assert torch.allclose(ga, a2.grad, atol=1e-4), \
         (ga - a2.grad).abs().max().item()
+
+
+def test_segment_mean_bf16_matches_fp32():
+    """bf16 segment mean: fp32 accumulate inside, only stores round —
+    compare against the plain fp32 torch reference with bf16 tolerance."""
+    torch.manual_seed(0)
+    n_src, n_dst, dim = 5000, 700, 100
+    x32 = torch.randn(n_src, dim, device="cuda")
+    x16 = x32.to(torch.bfloat16)
+    deg = torch.randint(1, 30, (n_dst,), device="cuda")
+    dst_ptr = torch.zeros(n_dst + 1, dtype=torch.long, device="cuda")
+    torch.cumsum(deg, 0, out=dst_ptr[1:])
+    E = int(dst_ptr[-1])
+    src = torch.randint(0, n_src, (E,), device="cuda")
+    out16 = _ext.segment_mean_gather(x16, src, dst_ptr)
+    assert out16.dtype == torch.bfloat16
+    # fp32 reference
+    ref = torch.zeros(n_dst, dim, device="cuda")
+    dst = torch.repeat_interleave(torch.arange(n_dst, device="cuda"), deg)
+    ref.index_add_(0, dst, x16.float()[src])
+    ref = ref / deg.clamp(min=1).unsqueeze(-1).float()
+    assert torch.allclose(out16.float(), ref, atol=3e-2, rtol=3e-2)
+
+    # backward: packed-bf16 atomics vs fp32 scatter reference
+    g16 = torch.randn(n_dst, dim, device="cuda").to(torch.bfloat16)
+    gx = _ext.segment_mean_gather_backward(g16, src, dst_ptr, n_src)
+    assert gx.dtype == torch.bfloat16
+    ref_gx = torch.zeros(n_src, dim, device="cuda")
+    contrib = (g16.float() / deg.clamp(min=1).unsqueeze(-1).float())[dst]
+    ref_gx.index_add_(0, src, contrib)
+    assert torch.allclose(gx.float(), ref_gx, atol=6e-2, rtol=6e-2)
+
+
+def test_sage_bf16_e2e_step():
+    """One fwd+bwd of bf16 GraphSAGE through the fused kernels."""
+    from quiver.nn import GraphSAGE
+    import quiver
+    g = torch.Generator().manual_seed(0)
+    n = 3000
+    deg = torch.randint(1, 20, (n,), generator=g)
+    indptr = torch.zeros(n + 1, dtype=torch.long)
+    torch.cumsum(deg, 0, out=indptr[1:])
+    indices = torch.randint(0, n, (int(indptr[-1]),), generator=g)
+    topo = quiver.CSRTopo(indptr=indptr, indices=indices)
+    s = quiver.GraphSageSampler(topo, [8, 4], device=0, mode="GPU")
+    n_id, bs, adjs = s.sample(torch.arange(64))
+    x = torch.randn(n_id.numel(), 100, generator=g) \
+        .to(torch.bfloat16).cuda().requires_grad_(True)
+    model = GraphSAGE(100, 64, 10, num_layers=2, dropout=0.0) \
+        .to(torch.bfloat16).cuda()
+    out = model(x, adjs)
+    assert out.dtype == torch.bfloat16
+    loss = torch.nn.functional.nll_loss(
+        out.float(), torch.randint(0, 10, (bs,), device="cuda"))
+    loss.backward()
+    assert x.grad is not None and torch.isfinite(x.grad.float()).all()
+    for p in model.parameters():
+        assert p.grad is None or torch.isfinite(p.grad.float()).all()
