@@ -3,6 +3,7 @@ import pytest
 import torch
 
 import quiver
+from quiver import _ext
 from quiver.nn import _mean_aggregate, GraphSAGE
 
 pytestmark = pytest.mark.gpu
