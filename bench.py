@@ -104,6 +104,18 @@ def main():
                    help="disable the sample+gather / compute prefetch overlap")
     p.add_argument("--prefetch-streams", type=int, default=1)
     p.add_argument("--model", default="sage", choices=["sage", "gat"])
+    p.add_argument("--placement", default="degree",
+                   choices=["degree", "prob"],
+                   help="hot-cache ordering: out-degree (default) or "
+                        "multi-hop access probability from sample_prob "
+                        "(reference cal_neighbor_prob).  On THIS synthetic "
+                        "graph destination hotness is degree rank by "
+                        "construction, so both orders coincide; the flag "
+                        "exists for real datasets where they do not")
+    p.add_argument("--report-hit-rate", action="store_true",
+                   help="measure the hot-cache hit rate over the warmup "
+                        "batches (fraction of gathered rows served from "
+                        "HBM) and print it before the timed region")
     p.add_argument("--dtype", default="fp32", choices=["fp32", "bf16"],
                    help="feature + model compute dtype.  The headline "
                         "config is fp32 (the reference's); bf16 halves "
@@ -179,7 +191,8 @@ def main():
                 dist.all_gather_object(objs, obj)
                 return objs
 
-            f.from_cpu_tensor_dist(feat_cpu, world, rank, all_gather)
+            f.from_cpu_tensor_dist(feat_cpu, world, rank, all_gather,
+                                   score=placement_score)
             return f
         if policy == "p2p_clique_replicate":
             quiver.init_p2p(list(range(world)))
@@ -189,11 +202,18 @@ def main():
         f = quiver.Feature(local_rank, device_list=devices,
                            device_cache_size=args.cache,
                            cache_policy=policy, csr_topo=csr_topo)
-        f.from_cpu_tensor(feat_cpu)
+        f.from_cpu_tensor(feat_cpu, score=placement_score)
         return f
 
     sampler = quiver.GraphSageSampler(csr_topo, FANOUT, device=local_rank,
                                       mode=args.mode)
+    placement_score = None
+    if args.placement == "prob":
+        # multi-hop access probability from the training seed distribution
+        # (uniform over train_idx) drives the hot/cold ordering
+        tg0 = torch.Generator().manual_seed(42)
+        ti = torch.randint(0, args.nodes, (n_train,), generator=tg0)
+        placement_score = sampler.sample_prob(ti.to(device), args.nodes).cpu()
     try:
         feature = build_feature(args.cache_policy)
     except (RuntimeError, AssertionError) as e:
@@ -258,6 +278,22 @@ def main():
                 total += train_on(n_id, bs, adjs, x)
         return total
 
+    if args.report_hit_rate:
+        # fraction of gathered rows served from the HBM hot cache
+        row_bytes = feat_dim * feat_cpu.element_size()
+        hot_rows = quiver.utils.parse_size(args.cache) // row_bytes
+        if args.cache_policy == "p2p_clique_replicate":
+            hot_rows *= world
+        hits = tot = 0
+        for i in range(args.warmup):
+            n_id, _, _ = sampler.sample(batches[i])
+            rows = feature.feature_order[n_id] \
+                if feature.feature_order is not None else n_id
+            hits += int((rows < hot_rows).sum())
+            tot += n_id.numel()
+        print(f"# hot-cache hit rate ({args.placement} placement): "
+              f"{hits/max(tot,1):.4f} over {tot} gathered rows", flush=True)
+
     run_range(0, args.warmup)
     if distributed:
         dist.barrier()
@@ -315,6 +351,7 @@ def main():
                 "sample_mode": args.mode,
                 "cache": args.cache,
                 "cache_policy": args.cache_policy,
+                "placement": args.placement,
                 "overlap": not args.no_overlap,
                 "epoch_seconds_derived": epoch_seconds,
                 "ref_epoch_seconds": ref,
