@@ -141,3 +141,43 @@ def test_mixed_sampler_gpu(small_graph):
     for n_id, bs, adjs in results:
         assert bs == 20
         assert len(adjs) == 2
+
+
+@pytest.mark.parametrize("mode", ["GPU_CPU_MIXED", "UVA_CPU_MIXED"])
+def test_mixed_sampler_gpu_cpu_work_stealing(small_graph, mode):
+    """MIXED modes on real hardware: device sampler + CPU worker pool
+    split one epoch adaptively; every job batch must come back exactly
+    once with a valid computational graph."""
+    indptr, indices = small_graph
+    topo = quiver.CSRTopo(indptr=indptr, indices=indices)
+
+    class Job(quiver.SampleJob):
+        def __init__(self):
+            self.data = [torch.arange(i * 25, i * 25 + 25)
+                         for i in range(12)]
+
+        def __getitem__(self, i):
+            return self.data[i]
+
+        def __len__(self):
+            return len(self.data)
+
+        def shuffle(self):
+            pass
+
+    sampler = quiver.MixedGraphSageSampler(Job(), 2, topo, [6, 4],
+                                           device=0, mode=mode)
+    results = list(iter(sampler))
+    assert len(results) == 12
+    seen_first = set()
+    for n_id, bs, adjs in results:
+        assert bs == 25
+        assert len(adjs) == 2
+        n_id = n_id.cpu()
+        seen_first.add(int(n_id[0]))
+        for adj in adjs:
+            ei = adj.edge_index.cpu()
+            assert ei[0].max() < adj.size[0]
+            assert ei[1].max() < adj.size[1]
+    assert len(seen_first) == 12  # every batch exactly once
+    sampler.shutdown()
