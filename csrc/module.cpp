@@ -1327,6 +1327,53 @@ std::tuple<torch::Tensor, torch::Tensor> gat_alpha_backward(
     return {g_asrc, g_adst};
 }
 
+// Fused GAT attention dots (see segment_kernels.hip).
+std::tuple<torch::Tensor, torch::Tensor> gat_dots(torch::Tensor h,
+                                                  torch::Tensor att_src,
+                                                  torch::Tensor att_dst,
+                                                  int64_t n_dst,
+                                                  int64_t heads) {
+    TORCH_CHECK(h.is_cuda() && h.dtype() == torch::kFloat32 && h.dim() == 2);
+    h = h.contiguous();
+    att_src = att_src.contiguous();
+    att_dst = att_dst.contiguous();
+    const int64_t n = h.size(0);
+    const int chead = (int)(h.size(1) / heads);
+    TORCH_CHECK((int64_t)chead * heads == h.size(1), "dim % heads != 0");
+    TORCH_CHECK(att_src.numel() == h.size(1) &&
+                att_dst.numel() == h.size(1));
+    auto asrc = torch::empty({n, heads}, h.options());
+    auto adst = torch::empty({n_dst, heads}, h.options());
+    qk::launch_gat_dots_fwd(current_stream(), h.data_ptr<float>(),
+                            att_src.data_ptr<float>(),
+                            att_dst.data_ptr<float>(), n, n_dst, (int)heads,
+                            chead, asrc.data_ptr<float>(),
+                            adst.data_ptr<float>());
+    return {asrc, adst};
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> gat_dots_backward(
+    torch::Tensor h, torch::Tensor att_src, torch::Tensor att_dst,
+    torch::Tensor g_asrc, torch::Tensor g_adst, int64_t heads) {
+    h = h.contiguous();
+    att_src = att_src.contiguous();
+    att_dst = att_dst.contiguous();
+    g_asrc = g_asrc.contiguous();
+    g_adst = g_adst.contiguous();
+    const int64_t n = h.size(0), n_dst = g_adst.size(0);
+    const int chead = (int)(h.size(1) / heads);
+    auto g_h = torch::empty_like(h);
+    auto g_as = torch::zeros_like(att_src);
+    auto g_ad = torch::zeros_like(att_dst);
+    qk::launch_gat_dots_bwd(
+        current_stream(), h.data_ptr<float>(), att_src.data_ptr<float>(),
+        att_dst.data_ptr<float>(), g_asrc.data_ptr<float>(),
+        g_adst.data_ptr<float>(), n, n_dst, (int)heads, chead,
+        g_h.data_ptr<float>(), g_as.data_ptr<float>(),
+        g_ad.data_ptr<float>());
+    return {g_h, g_as, g_ad};
+}
+
 void init_p2p(const std::vector<int>& devices) {
     // On an 8x MI355X node every pair is xGMI-connected: enable the full
     // clique (reference init_p2p, quiver_feature.cu:378-421; no NVLink-style
@@ -1513,6 +1560,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("segment_mean_gather_backward", &segment_mean_gather_backward,
           py::call_guard<py::gil_scoped_release>());
 
+    m.def("gat_dots", &gat_dots,
+          "fused per-head attention dots over a projected frontier");
+    m.def("gat_dots_backward", &gat_dots_backward);
     m.def("gat_alpha", &gat_alpha,
           "fused GAT attention coefficients over dst-sorted edges");
     m.def("gat_alpha_backward", &gat_alpha_backward);

@@ -138,6 +138,19 @@ void launch_gather(hipStream_t s, const GatherSpec& spec,
 void launch_scatter(hipStream_t s, const GatherSpec& spec,
                     const int64_t* indices, int64_t n, const char* src);
 
+// Fused GAT attention dots: asrc[n,h] = <h[n,h,:], att_src[h,:]>, adst
+// over the first n_dst rows (prefix convention); backward also emits
+// g_h and the (tiny) att gradients via per-block LDS accumulation.
+void launch_gat_dots_fwd(hipStream_t s, const float* h,
+                         const float* att_src, const float* att_dst,
+                         int64_t n, int64_t n_dst, int heads, int chead,
+                         float* asrc, float* adst);
+void launch_gat_dots_bwd(hipStream_t s, const float* h,
+                         const float* att_src, const float* att_dst,
+                         const float* g_asrc, const float* g_adst,
+                         int64_t n, int64_t n_dst, int heads, int chead,
+                         float* g_h, float* g_att_src, float* g_att_dst);
+
 // ---------- tall-skinny weight-grad GEMM (wgrad_kernels.hip) -------------
 
 // C[M×N] = A^T @ B with A [K×M], B [K×N] row-major, K huge.  MFMA

@@ -392,6 +392,118 @@ gat_alpha_bwd_kernel(const float* __restrict__ grad_alpha,
     }
 }
 
+// ---- fused GAT attention dots --------------------------------------------
+// asrc[n,h] = <h[n,h,:], att_src[h,:]>; adst likewise over the first
+// n_dst rows (bipartite prefix convention: h_dst == h_src[:n_dst]).
+// Replaces the torch chain (h*att).sum(-1) x2 (a 1 GB broadcast-mul
+// write + 1 GB reduce read per term) with ONE read of h.
+constexpr int DSUB = 16;  // lanes per (row, head) pair
+
+__global__ void __launch_bounds__(BLOCK)
+gat_dots_fwd_kernel(const float* __restrict__ h,
+                    const float* __restrict__ att_src,
+                    const float* __restrict__ att_dst, int64_t n,
+                    int64_t n_dst, int heads, int chead,
+                    float* __restrict__ asrc, float* __restrict__ adst) {
+    const int sub_id = threadIdx.x / DSUB;
+    const int lane = threadIdx.x % DSUB;
+    const int per_block = BLOCK / DSUB;
+    int64_t p = (int64_t)blockIdx.x * per_block + sub_id;
+    const int64_t stride = (int64_t)gridDim.x * per_block;
+    const int64_t total = n * heads;
+    for (; p < total; p += stride) {
+        const int64_t row = p / heads;
+        const int hd = (int)(p % heads);
+        const float* hrow = h + (row * heads + hd) * (int64_t)chead;
+        const float* as = att_src + hd * chead;
+        const float* ad = att_dst + hd * chead;
+        float sv = 0.f, dv = 0.f;
+        const bool do_dst = row < n_dst;
+        for (int c = lane * 4; c < chead; c += DSUB * 4) {
+            const float4 v = *reinterpret_cast<const float4*>(hrow + c);
+            const float4 a = *reinterpret_cast<const float4*>(as + c);
+            sv += v.x * a.x + v.y * a.y + v.z * a.z + v.w * a.w;
+            if (do_dst) {
+                const float4 b = *reinterpret_cast<const float4*>(ad + c);
+                dv += v.x * b.x + v.y * b.y + v.z * b.z + v.w * b.w;
+            }
+        }
+        for (int off = DSUB / 2; off; off >>= 1) {
+            sv += __shfl_down(sv, off, DSUB);
+            dv += __shfl_down(dv, off, DSUB);
+        }
+        if (lane == 0) {
+            asrc[row * heads + hd] = sv;
+            if (do_dst) adst[row * heads + hd] = dv;
+        }
+    }
+}
+
+// g_h[row,h,:] = g_asrc[row,h]*att_src[h,:] (+ g_adst part for prefix
+// rows); g_att_* accumulate per-block in LDS, one global atomic per
+// element per block (the [H*C] output is tiny next to the 1M-row input).
+__global__ void __launch_bounds__(BLOCK)
+gat_dots_bwd_kernel(const float* __restrict__ h,
+                    const float* __restrict__ att_src,
+                    const float* __restrict__ att_dst,
+                    const float* __restrict__ g_asrc,
+                    const float* __restrict__ g_adst, int64_t n,
+                    int64_t n_dst, int heads, int chead,
+                    float* __restrict__ g_h, float* __restrict__ g_att_src,
+                    float* __restrict__ g_att_dst) {
+    extern __shared__ float lacc[];  // [2][heads*chead]
+    const int D = heads * chead;
+    float* lsrc = lacc;
+    float* ldst = lacc + D;
+    for (int i = threadIdx.x; i < 2 * D; i += BLOCK) lacc[i] = 0.f;
+    __syncthreads();
+    const int sub_id = threadIdx.x / DSUB;
+    const int lane = threadIdx.x % DSUB;
+    const int per_block = BLOCK / DSUB;
+    int64_t p = (int64_t)blockIdx.x * per_block + sub_id;
+    const int64_t stride = (int64_t)gridDim.x * per_block;
+    const int64_t total = n * heads;
+    for (; p < total; p += stride) {
+        const int64_t row = p / heads;
+        const int hd = (int)(p % heads);
+        const float* hrow = h + (row * heads + hd) * (int64_t)chead;
+        float* grow = g_h + (row * heads + hd) * (int64_t)chead;
+        const float* as = att_src + hd * chead;
+        const float* ad = att_dst + hd * chead;
+        const bool do_dst = row < n_dst;
+        const float gs = g_asrc[row * heads + hd];
+        const float gd = do_dst ? g_adst[row * heads + hd] : 0.f;
+        for (int c = lane * 4; c < chead; c += DSUB * 4) {
+            const float4 v = *reinterpret_cast<const float4*>(hrow + c);
+            const float4 a = *reinterpret_cast<const float4*>(as + c);
+            float4 g;
+            g.x = gs * a.x; g.y = gs * a.y; g.z = gs * a.z; g.w = gs * a.w;
+            if (do_dst) {
+                const float4 b = *reinterpret_cast<const float4*>(ad + c);
+                g.x += gd * b.x; g.y += gd * b.y;
+                g.z += gd * b.z; g.w += gd * b.w;
+            }
+            *reinterpret_cast<float4*>(grow + c) = g;
+            const int base = hd * chead + c;
+            atomicAdd(&lsrc[base + 0], gs * v.x);
+            atomicAdd(&lsrc[base + 1], gs * v.y);
+            atomicAdd(&lsrc[base + 2], gs * v.z);
+            atomicAdd(&lsrc[base + 3], gs * v.w);
+            if (do_dst) {
+                atomicAdd(&ldst[base + 0], gd * v.x);
+                atomicAdd(&ldst[base + 1], gd * v.y);
+                atomicAdd(&ldst[base + 2], gd * v.z);
+                atomicAdd(&ldst[base + 3], gd * v.w);
+            }
+        }
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < D; i += BLOCK) {
+        atomicAdd(&g_att_src[i], lsrc[i]);
+        atomicAdd(&g_att_dst[i], ldst[i]);
+    }
+}
+
 inline int grid_for(int64_t work, int per_block) {
     int64_t blocks = (work + per_block - 1) / per_block;
     if (blocks > 2048) blocks = 2048;
@@ -522,6 +634,37 @@ void launch_gat_alpha_bwd(hipStream_t s, const float* grad_alpha,
     gat_alpha_bwd_kernel<<<grid_for(n_dst * heads, BLOCK), BLOCK, 0, s>>>(
         grad_alpha, alpha, asrc, adst, src, dst_ptr, n_dst, heads, slope,
         g_asrc, g_adst);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+
+void launch_gat_dots_fwd(hipStream_t s, const float* h,
+                         const float* att_src, const float* att_dst,
+                         int64_t n, int64_t n_dst, int heads, int chead,
+                         float* asrc, float* adst) {
+    if (n == 0) return;
+    if (chead % 4 != 0)
+        throw std::runtime_error("gat_dots: chead must be a multiple of 4");
+    gat_dots_fwd_kernel<<<grid_for(n * heads, BLOCK / DSUB), BLOCK, 0, s>>>(
+        h, att_src, att_dst, n, n_dst, heads, chead, asrc, adst);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+void launch_gat_dots_bwd(hipStream_t s, const float* h,
+                         const float* att_src, const float* att_dst,
+                         const float* g_asrc, const float* g_adst,
+                         int64_t n, int64_t n_dst, int heads, int chead,
+                         float* g_h, float* g_att_src, float* g_att_dst) {
+    if (n == 0) return;
+    if (chead % 4 != 0)
+        throw std::runtime_error("gat_dots: chead must be a multiple of 4");
+    const int D = heads * chead;
+    if (D > 4096)
+        throw std::runtime_error("gat_dots: heads*chead too large for LDS");
+    gat_dots_bwd_kernel<<<grid_for(n * heads, BLOCK / DSUB), BLOCK,
+                          2 * D * sizeof(float), s>>>(
+        h, att_src, att_dst, g_asrc, g_adst, n, n_dst, heads, chead, g_h,
+        g_att_src, g_att_dst);
     QK_CHECK_HIP(hipGetLastError());
 }
 
