@@ -23,6 +23,7 @@ KERNEL_SOURCES = [
     "gather_kernels.hip",
     "segment_kernels.hip",
     "wgrad_kernels.hip",
+    "gemm_kernels.hip",
 ]
 TORCH_SOURCES = ["module.cpp"]
 OUT = os.path.join(ROOT, "torch_quiver.so")

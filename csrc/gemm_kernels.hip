@@ -1,0 +1,173 @@
+// Tall-M GEMM for gfx950: C[M×N] = A[M×K] @ B[K×N] (+ bias[N]) where M is
+// a GNN frontier (10^5..10^6 rows) and K, N are layer widths (<= 1024).
+//
+// This is the OTHER hot GEMM family of the model layer (wgrad_kernels.hip
+// covers A^T@B): the GAT projection forward runs x[1.06M×100] @ W^T and
+// its data-grad — rocBLAS picks MT256x64 stream-K tiles and lands ~5.8x
+// off the memory floor (1.12 ms for a read-424MB/write-1.08GB op).
+// MI355X-first design:
+//  - v_mfma_f32_32x32x2_f32 tiles (exact f32), one 32x32 tile per wave,
+//    2x2 waves = a 64x64 C macro-tile per block.  No split-K: M/64 tiles
+//    give tens of thousands of blocks by themselves.
+//  - B is staged K-major (the caller passes W^T for forward — a 100 KB
+//    host-side transpose — and W itself for data-grad, so both read
+//    coalesced), A staged through LDS with a +1 row pad (fragment lanes
+//    read a k-column of 32 consecutive rows: odd pitch -> conflict-free).
+//  - Double-buffered through registers AND ping-pong LDS, same skeleton
+//    as wgrad.
+#include "qk_common.h"
+
+namespace qk {
+
+namespace {
+
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+constexpr int BM = 64;
+constexpr int BN = 64;
+constexpr int BK = 32;
+constexpr int AP = BK + 1;   // As row pitch (odd -> fragment reads hit
+                             // distinct banks across the 32-row group)
+constexpr int BP = BN + 4;   // Bs row pitch
+constexpr int NT = 256;      // 4 waves
+
+// A slice: rows [m0, m0+BM) x k [k0, k0+BK); thread t loads 2 float4
+// (BM*BK/4/NT) at flat float4 index f = t + j*NT -> (row = f/8, kk4 =
+// f%8) covering k = kk4*4..+3.
+__device__ __forceinline__ void load_a(const float* __restrict__ A,
+                                       int64_t M, int K, int64_t m0, int k0,
+                                       int tid, float4 r[2]) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+        const int f = tid + j * NT;
+        const int64_t row = m0 + f / 8;
+        const int k = k0 + (f % 8) * 4;
+        if (row < M && k + 3 < K && (K % 4 == 0)) {
+            r[j] = *reinterpret_cast<const float4*>(&A[row * K + k]);
+        } else {
+            float v[4];
+#pragma unroll
+            for (int q = 0; q < 4; ++q)
+                v[q] = (row < M && k + q < K) ? A[row * K + k + q] : 0.f;
+            r[j] = make_float4(v[0], v[1], v[2], v[3]);
+        }
+    }
+}
+
+__device__ __forceinline__ void store_a(float (*As)[AP], int tid,
+                                        const float4 r[2]) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+        const int f = tid + j * NT;
+        const int row = f / 8, k = (f % 8) * 4;
+        As[row][k + 0] = r[j].x;
+        As[row][k + 1] = r[j].y;
+        As[row][k + 2] = r[j].z;
+        As[row][k + 3] = r[j].w;
+    }
+}
+
+// B slice: k [k0, k0+BK) x n [n0, n0+BN); thread t loads 2 float4 at
+// flat index f = t + j*NT -> (kk = f/16, n4 = f%16).
+__device__ __forceinline__ void load_b(const float* __restrict__ B, int K,
+                                       int N, int k0, int n0, int tid,
+                                       float4 r[2]) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+        const int f = tid + j * NT;
+        const int kk = k0 + f / 16;
+        const int n = n0 + (f % 16) * 4;
+        if (kk < K && n + 3 < N && (N % 4 == 0)) {
+            r[j] = *reinterpret_cast<const float4*>(&B[(int64_t)kk * N + n]);
+        } else {
+            float v[4];
+#pragma unroll
+            for (int q = 0; q < 4; ++q)
+                v[q] = (kk < K && n + q < N) ? B[(int64_t)kk * N + n + q]
+                                             : 0.f;
+            r[j] = make_float4(v[0], v[1], v[2], v[3]);
+        }
+    }
+}
+
+__device__ __forceinline__ void store_b(float (*Bs)[BP], int tid,
+                                        const float4 r[2]) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+        const int f = tid + j * NT;
+        *reinterpret_cast<float4*>(&Bs[f / 16][(f % 16) * 4]) = r[j];
+    }
+}
+
+__global__ void __launch_bounds__(NT)
+tall_gemm_kernel(const float* __restrict__ A, const float* __restrict__ B,
+                 const float* __restrict__ bias, float* __restrict__ C,
+                 int64_t M, int K, int N) {
+    __shared__ float As[2][BM][AP];
+    __shared__ float Bs[2][BK][BP];
+
+    const int64_t m0 = (int64_t)blockIdx.x * BM;
+    const int n0 = blockIdx.y * BN;
+    const int tid = threadIdx.x;
+    const int wave = tid / 64;
+    const int lane = tid % 64;
+    const int wm = (wave % 2) * 32;
+    const int wn = (wave / 2) * 32;
+    const int fcol = lane % 32;
+    const int fk = lane / 32;
+
+    f32x16 acc = {};
+    float4 ra[2], rb[2];
+    load_a(A, M, K, m0, 0, tid, ra);
+    load_b(B, K, N, 0, n0, tid, rb);
+    store_a(As[0], tid, ra);
+    store_b(Bs[0], tid, rb);
+    __syncthreads();
+
+    int buf = 0;
+    for (int k0 = 0; k0 < K; k0 += BK) {
+        const bool more = k0 + BK < K;
+        if (more) {
+            load_a(A, M, K, m0, k0 + BK, tid, ra);
+            load_b(B, K, N, k0 + BK, n0, tid, rb);
+        }
+#pragma unroll
+        for (int kk = 0; kk < BK; kk += 2) {
+            // A fragment: lane (col=fcol -> C row m, k=fk); the odd AP
+            // pitch spreads the 32-row column read across banks
+            const float a = As[buf][wm + fcol][kk + fk];
+            const float b = Bs[buf][kk + fk][wn + fcol];
+            acc = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc, 0, 0, 0);
+        }
+        if (more) {
+            store_a(As[buf ^ 1], tid, ra);
+            store_b(Bs[buf ^ 1], tid, rb);
+        }
+        __syncthreads();
+        buf ^= 1;
+    }
+
+    // D mapping (same orientation as the wgrad kernel: the first operand
+    // holds the LEFT matrix's rows): row m_out = (r&3)+8*(r>>2)+4*fk,
+    // col n_out = fcol — lanes store consecutive n, coalesced.
+    const int n_out = n0 + wn + fcol;
+    if (n_out >= N) return;
+    const float badd = bias ? bias[n_out] : 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+        const int64_t m_out = m0 + wm + (r & 3) + 8 * (r >> 2) + 4 * fk;
+        if (m_out < M) C[m_out * N + n_out] = acc[r] + badd;
+    }
+}
+
+}  // namespace
+
+void launch_tall_gemm(hipStream_t s, const float* A, const float* B,
+                      const float* bias, float* C, int64_t M, int K, int N) {
+    if (M == 0 || K == 0 || N == 0) return;
+    dim3 grid((unsigned)((M + BM - 1) / BM), (unsigned)((N + BN - 1) / BN));
+    tall_gemm_kernel<<<grid, NT, 0, s>>>(A, B, bias, C, M, K, N);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+}  // namespace qk

@@ -1327,6 +1327,33 @@ std::tuple<torch::Tensor, torch::Tensor> gat_alpha_backward(
     return {g_asrc, g_adst};
 }
 
+// Tall-M GEMM: out[M×N] = a[M×K] @ b_kmajor[K×N] (+bias).  The caller
+// passes W^T (tiny one-off transpose) for a forward linear and W itself
+// for the data-grad, so both read coalesced.
+torch::Tensor tall_gemm(torch::Tensor a, torch::Tensor b_kmajor,
+                        torch::Tensor bias) {
+    TORCH_CHECK(a.is_cuda() && a.dim() == 2 &&
+                a.dtype() == torch::kFloat32 &&
+                b_kmajor.dtype() == torch::kFloat32 &&
+                a.size(1) == b_kmajor.size(0),
+                "tall_gemm: fp32 [M,K] @ [K,N] expected");
+    a = a.contiguous();
+    b_kmajor = b_kmajor.contiguous();
+    const int64_t m = a.size(0);
+    const int k = (int)a.size(1), n = (int)b_kmajor.size(1);
+    auto c = torch::empty({m, (int64_t)n}, a.options());
+    const float* bp = nullptr;
+    if (bias.defined() && bias.numel() > 0) {
+        bias = bias.contiguous();
+        TORCH_CHECK(bias.numel() == n, "bias size mismatch");
+        bp = bias.data_ptr<float>();
+    }
+    qk::launch_tall_gemm(current_stream(), a.data_ptr<float>(),
+                         b_kmajor.data_ptr<float>(), bp,
+                         c.data_ptr<float>(), m, k, n);
+    return c;
+}
+
 // Fused GAT attention dots (see segment_kernels.hip).
 std::tuple<torch::Tensor, torch::Tensor> gat_dots(torch::Tensor h,
                                                   torch::Tensor att_src,
@@ -1560,6 +1587,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("segment_mean_gather_backward", &segment_mean_gather_backward,
           py::call_guard<py::gil_scoped_release>());
 
+    m.def("tall_gemm", &tall_gemm,
+          "C[M,N] = A[M,K] @ B[K,N] (+bias), M huge, MFMA 32x32 tiles");
     m.def("gat_dots", &gat_dots,
           "fused per-head attention dots over a projected frontier");
     m.def("gat_dots_backward", &gat_dots_backward);

@@ -151,6 +151,13 @@ void launch_gat_dots_bwd(hipStream_t s, const float* h,
                          int64_t n, int64_t n_dst, int heads, int chead,
                          float* g_h, float* g_att_src, float* g_att_dst);
 
+// ---------- tall-M GEMM (gemm_kernels.hip) -------------------------------
+
+// C[M×N] = A[M×K] @ B[K×N] (+ bias[N]) with M huge, K/N <= ~1024 and B
+// K-major (pass W^T for a forward linear, W itself for its data-grad).
+void launch_tall_gemm(hipStream_t s, const float* A, const float* B,
+                      const float* bias, float* C, int64_t M, int K, int N);
+
 // ---------- tall-skinny weight-grad GEMM (wgrad_kernels.hip) -------------
 
 // C[M×N] = A^T @ B with A [K×M], B [K×N] row-major, K huge.  MFMA

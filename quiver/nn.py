@@ -137,25 +137,29 @@ class _QLinearFn(torch.autograd.Function):
     """Linear layer whose weight gradient uses the split-K HIP kernel
     (csrc/wgrad_kernels.hip).
 
-    The wgrad GEMM here is tall-skinny — C[out×in] = grad_out^T @ x with
-    K = frontier rows (10^4..10^6) — a shape rocBLAS stream-K runs ~60x
-    off the HBM roof on gfx950.  Forward and data-grad GEMMs keep rocBLAS
-    (their shapes are fine).  Bias grad is folded into the same kernel
-    pass.  fp32 atomic split-K: wgrad is bitwise-nondeterministic (same
-    class of nondeterminism as rocBLAS GSU).
+    All three GEMMs of a big-frontier linear run on custom MFMA kernels:
+    - forward / data-grad: tall-M kernel (csrc/gemm_kernels.hip) — rocBLAS
+      runs the 1M-row GAT projection ~5.8x off the memory floor;
+    - weight-grad: split-K tall-skinny kernel (csrc/wgrad_kernels.hip,
+      deterministic workspace reduce) with the bias grad folded in.
+    Numerics are exact f32 fmaf chains (v_mfma_f32_32x32x2_f32).
     """
 
     @staticmethod
     def forward(ctx, x, weight, bias):
         ctx.save_for_backward(x, weight)
         ctx.has_bias = bias is not None
-        return F.linear(x, weight, bias)
+        # forward on the tall-M MFMA kernel: B is W^T (a tiny one-off
+        # transpose) so the kernel reads K-major coalesced
+        return _ext.tall_gemm(x, weight.t().contiguous(), bias)
 
     @staticmethod
     def backward(ctx, grad_out):
         x, weight = ctx.saved_tensors
         grad_out = grad_out.contiguous()
-        grad_x = grad_out @ weight if ctx.needs_input_grad[0] else None
+        # data-grad reads W itself K-major: same kernel, no transpose
+        grad_x = _ext.tall_gemm(grad_out, weight, None) \
+            if ctx.needs_input_grad[0] else None
         grad_w, grad_b = _ext.wgrad(grad_out, x, ctx.has_bias)
         return grad_x, grad_w, (grad_b if ctx.has_bias else None)
 
