@@ -106,7 +106,13 @@ tall_gemm_kernel(const float* __restrict__ A, const float* __restrict__ B,
     __shared__ float As[2][BM][AP];
     __shared__ float Bs[2][BK][BP];
 
-    const int64_t m0 = (int64_t)blockIdx.x * BM;
+    // m grid-stride: with K <= a few hundred a single 64x64 tile is only
+    // 2-8 k-stages — block setup and pipeline fill dominate (measured
+    // 0.5-0.9x rocBLAS as a one-tile-per-block kernel).  Each block
+    // walks several m-tiles with one CONTINUOUS ping-pong pipeline: the
+    // next tile's first slices prefetch during the current tile's last
+    // k-stage, and the small B panel stays L2-hot across tiles.
+    const int64_t m_stride = (int64_t)gridDim.x * BM;
     const int n0 = blockIdx.y * BN;
     const int tid = threadIdx.x;
     const int wave = tid / 64;
@@ -116,8 +122,11 @@ tall_gemm_kernel(const float* __restrict__ A, const float* __restrict__ B,
     const int fcol = lane % 32;
     const int fk = lane / 32;
 
-    f32x16 acc = {};
+    const int n_out = n0 + wn + fcol;
+    const float badd = (bias && n_out < N) ? bias[n_out] : 0.f;
+
     float4 ra[2], rb[2];
+    int64_t m0 = (int64_t)blockIdx.x * BM;
     load_a(A, M, K, m0, 0, tid, ra);
     load_b(B, K, N, 0, n0, tid, rb);
     store_a(As[0], tid, ra);
@@ -125,38 +134,45 @@ tall_gemm_kernel(const float* __restrict__ A, const float* __restrict__ B,
     __syncthreads();
 
     int buf = 0;
-    for (int k0 = 0; k0 < K; k0 += BK) {
-        const bool more = k0 + BK < K;
-        if (more) {
-            load_a(A, M, K, m0, k0 + BK, tid, ra);
-            load_b(B, K, N, k0 + BK, n0, tid, rb);
-        }
+    for (; m0 < M; m0 += m_stride) {
+        f32x16 acc = {};
+        for (int k0 = 0; k0 < K; k0 += BK) {
+            const bool more_k = k0 + BK < K;
+            const bool more_m = m0 + m_stride < M;
+            if (more_k) {
+                load_a(A, M, K, m0, k0 + BK, tid, ra);
+                load_b(B, K, N, k0 + BK, n0, tid, rb);
+            } else if (more_m) {
+                load_a(A, M, K, m0 + m_stride, 0, tid, ra);
+                load_b(B, K, N, 0, n0, tid, rb);
+            }
 #pragma unroll
-        for (int kk = 0; kk < BK; kk += 2) {
-            // A fragment: lane (col=fcol -> C row m, k=fk); the odd AP
-            // pitch spreads the 32-row column read across banks
-            const float a = As[buf][wm + fcol][kk + fk];
-            const float b = Bs[buf][kk + fk][wn + fcol];
-            acc = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc, 0, 0, 0);
+            for (int kk = 0; kk < BK; kk += 2) {
+                // A fragment: lane (row = fcol, k = fk); the odd AP
+                // pitch spreads the 32-row column read across banks
+                const float a = As[buf][wm + fcol][kk + fk];
+                const float b = Bs[buf][kk + fk][wn + fcol];
+                acc = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc,
+                                                           0, 0, 0);
+            }
+            if (more_k || more_m) {
+                store_a(As[buf ^ 1], tid, ra);
+                store_b(Bs[buf ^ 1], tid, rb);
+            }
+            __syncthreads();
+            buf ^= 1;
         }
-        if (more) {
-            store_a(As[buf ^ 1], tid, ra);
-            store_b(Bs[buf ^ 1], tid, rb);
-        }
-        __syncthreads();
-        buf ^= 1;
-    }
-
-    // D mapping (same orientation as the wgrad kernel: the first operand
-    // holds the LEFT matrix's rows): row m_out = (r&3)+8*(r>>2)+4*fk,
-    // col n_out = fcol — lanes store consecutive n, coalesced.
-    const int n_out = n0 + wn + fcol;
-    if (n_out >= N) return;
-    const float badd = bias ? bias[n_out] : 0.f;
+        // D mapping (same orientation as the wgrad kernel: the first
+        // operand holds the LEFT matrix's rows): row m_out =
+        // (r&3)+8*(r>>2)+4*fk, col n_out = fcol — coalesced stores.
+        if (n_out < N) {
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-        const int64_t m_out = m0 + wm + (r & 3) + 8 * (r >> 2) + 4 * fk;
-        if (m_out < M) C[m_out * N + n_out] = acc[r] + badd;
+            for (int r = 0; r < 16; ++r) {
+                const int64_t m_out =
+                    m0 + wm + (r & 3) + 8 * (r >> 2) + 4 * fk;
+                if (m_out < M) C[m_out * N + n_out] = acc[r] + badd;
+            }
+        }
     }
 }
 
@@ -165,7 +181,12 @@ tall_gemm_kernel(const float* __restrict__ A, const float* __restrict__ B,
 void launch_tall_gemm(hipStream_t s, const float* A, const float* B,
                       const float* bias, float* C, int64_t M, int K, int N) {
     if (M == 0 || K == 0 || N == 0) return;
-    dim3 grid((unsigned)((M + BM - 1) / BM), (unsigned)((N + BN - 1) / BN));
+    int64_t tm = (M + BM - 1) / BM;
+    int tn = (N + BN - 1) / BN;
+    // enough blocks to fill 256 CUs several times over, few enough that
+    // each walks multiple m-tiles (pipeline amortization)
+    int gx = (int)std::min<int64_t>(tm, 4096 / tn);
+    dim3 grid((unsigned)gx, (unsigned)tn);
     tall_gemm_kernel<<<grid, NT, 0, s>>>(A, B, bias, C, M, K, N);
     QK_CHECK_HIP(hipGetLastError());
 }
