@@ -25,6 +25,11 @@ def time_fn(fn, iters=50, warmup=10):
 
 
 def main():
+    import argparse
+    p = argparse.ArgumentParser()
+    p.add_argument("--shape", default=None, help="M,K,N single shape")
+    p.add_argument("--iters", type=int, default=50)
+    args = p.parse_args()
     shapes = [
         (1_060_000, 100, 256),  # GAT layer-1 projection (products frontier)
         (1_060_000, 256, 100),  # its data-grad
@@ -32,6 +37,8 @@ def main():
         (90_000, 256, 256),
         (13_000, 256, 47),      # last layer
     ]
+    if args.shape:
+        shapes = [tuple(int(x) for x in args.shape.split(","))]
     g = torch.Generator(device="cuda").manual_seed(0)
     print(f"{'M':>9} {'K':>4} {'N':>4} | {'mfma us':>8} {'torch us':>9} "
           f"{'speedup':>7} {'GB/s':>7} {'rel err':>9}")
@@ -40,8 +47,8 @@ def main():
         w = torch.randn(n, k, device="cuda", generator=g)  # torch layout
         bias = torch.randn(n, device="cuda", generator=g)
         wt = w.t().contiguous()
-        t_q = time_fn(lambda: _ext.tall_gemm(a, wt, bias))
-        t_t = time_fn(lambda: torch.nn.functional.linear(a, w, bias))
+        t_q = time_fn(lambda: _ext.tall_gemm(a, wt, bias), iters=args.iters)
+        t_t = time_fn(lambda: torch.nn.functional.linear(a, w, bias), iters=args.iters)
         c = _ext.tall_gemm(a, wt, bias)
         want = torch.nn.functional.linear(a, w, bias)
         rel = float((c - want).norm() / want.norm())

@@ -1331,7 +1331,9 @@ std::tuple<torch::Tensor, torch::Tensor> gat_alpha_backward(
 // passes W^T (tiny one-off transpose) for a forward linear and W itself
 // for the data-grad, so both read coalesced.
 torch::Tensor tall_gemm(torch::Tensor a, torch::Tensor b_kmajor,
-                        torch::Tensor bias) {
+                        c10::optional<torch::Tensor> bias_opt) {
+    torch::Tensor bias =
+        bias_opt.has_value() ? *bias_opt : torch::Tensor();
     TORCH_CHECK(a.is_cuda() && a.dim() == 2 &&
                 a.dtype() == torch::kFloat32 &&
                 b_kmajor.dtype() == torch::kFloat32 &&
