@@ -137,29 +137,38 @@ class _QLinearFn(torch.autograd.Function):
     """Linear layer whose weight gradient uses the split-K HIP kernel
     (csrc/wgrad_kernels.hip).
 
-    All three GEMMs of a big-frontier linear run on custom MFMA kernels:
-    - forward / data-grad: tall-M kernel (csrc/gemm_kernels.hip) — rocBLAS
-      runs the 1M-row GAT projection ~5.8x off the memory floor;
-    - weight-grad: split-K tall-skinny kernel (csrc/wgrad_kernels.hip,
-      deterministic workspace reduce) with the bias grad folded in.
-    Numerics are exact f32 fmaf chains (v_mfma_f32_32x32x2_f32).
+    Big-frontier linear GEMM routing (all measured, benchmarks/):
+    - weight-grad: custom MFMA split-K (csrc/wgrad_kernels.hip,
+      deterministic workspace reduce, 2.8-5.7x rocBLAS on these shapes)
+      with the bias grad folded in;
+    - narrow-output data-grads (in_features <= 128): custom tall-M MFMA
+      kernel (csrc/gemm_kernels.hip, 1.1-1.25x);
+    - forward and wide data-grads: rocBLAS (it wins there).
+    Custom-kernel numerics are exact f32 fmaf chains
+    (v_mfma_f32_32x32x2_f32).
     """
 
     @staticmethod
     def forward(ctx, x, weight, bias):
         ctx.save_for_backward(x, weight)
         ctx.has_bias = bias is not None
-        # forward on the tall-M MFMA kernel: B is W^T (a tiny one-off
-        # transpose) so the kernel reads K-major coalesced
-        return _ext.tall_gemm(x, weight.t().contiguous(), bias)
+        return F.linear(x, weight, bias)
 
     @staticmethod
     def backward(ctx, grad_out):
         x, weight = ctx.saved_tensors
         grad_out = grad_out.contiguous()
-        # data-grad reads W itself K-major: same kernel, no transpose
-        grad_x = _ext.tall_gemm(grad_out, weight, None) \
-            if ctx.needs_input_grad[0] else None
+        if ctx.needs_input_grad[0]:
+            # narrow-output data-grads (in_features <= 128) beat rocBLAS
+            # on the tall-M MFMA kernel (measured 1.1-1.25x; wide shapes
+            # lose ~15% — rocBLAS keeps those; benchmarks/bench_gemm.py)
+            if weight.size(1) <= 128 and grad_out.is_cuda \
+                    and grad_out.dtype == torch.float32:
+                grad_x = _ext.tall_gemm(grad_out, weight, None)
+            else:
+                grad_x = grad_out @ weight
+        else:
+            grad_x = None
         grad_w, grad_b = _ext.wgrad(grad_out, x, ctx.has_bias)
         return grad_x, grad_w, (grad_b if ctx.has_bias else None)
 
