@@ -65,18 +65,26 @@ inline int grid_for(int64_t work, int per_block, int max_blocks) {
     return (int)blocks;
 }
 
-inline int gather_max_blocks(const GatherSpec& spec) {
+inline int gather_max_blocks(const GatherSpec& spec, bool overlapped) {
     // A gather touching a pinned-host (zero-copy) shard is PCIe-latency
-    // bound: a few hundred blocks keep enough requests in flight to
-    // saturate the link, and capping the grid leaves CUs free so the
-    // prefetch pipeline's gather truly overlaps model kernels on the other
-    // stream.  Pure-HBM/xGMI gathers want the full chip.
+    // bound and its uncached reads poison co-resident kernels, so the
+    // grid is capped to leave CUs free.  Two profiles, both measured on
+    // the products bench (blocks -> ms/step: 24:10.8, 56:5.0, 96:3.64,
+    // 192:3.9, 640:4.3, 1280:4.4):
+    //  - overlapped (the async sample->gather chain, which by
+    //    construction runs concurrently with model compute): 96 blocks
+    //    — the knee where the PCIe link still saturates but CU-slot
+    //    contention stops hurting the compute stream;
+    //  - standalone (direct Feature[...] indexing): 640 blocks for full
+    //    link occupancy when nothing competes.
+    // Pure-HBM/xGMI gathers want the full chip either way.
     static int env_cap = [] {
         const char* e = getenv("QUIVER_GATHER_BLOCKS");
         return e ? atoi(e) : 0;
     }();
     if (env_cap > 0) return env_cap;
-    return spec.has_host_shard ? 640 : 2048;
+    if (!spec.has_host_shard) return 2048;
+    return overlapped ? 96 : 640;
 }
 
 template <typename VecT, bool SCATTER>
@@ -86,7 +94,9 @@ void dispatch_sub(hipStream_t s, const GatherSpec& spec,
     int64_t nvec = spec.row_bytes / (int64_t)sizeof(VecT);
     int sub = 4;
     while (sub < 64 && sub < nvec) sub *= 2;  // cover the row in ~1 pass
-    int grid = grid_for(n, BLOCK / sub, gather_max_blocks(spec));
+    // n_dev-driven gathers are the async chain: always compute-overlapped
+    int grid = grid_for(n, BLOCK / sub,
+                        gather_max_blocks(spec, n_dev != nullptr));
     switch (sub) {
 #define QK_CASE(S)                                                          \
     case S:                                                                 \
