@@ -197,7 +197,14 @@ class GraphSageSampler:
 
     def sample_prob(self, train_idx, total_node_count):
         """Multi-hop access probability of every node when seeding from
-        train_idx — drives access-probability feature placement."""
+        train_idx — drives access-probability feature placement.
+
+        The propagation formula (reference cal_next,
+        cuda_random.cu.hpp:71-104) reads each node's CSR row as BOTH its
+        out- and in-edges, i.e. it assumes a symmetrized graph — true for
+        the OGB/Reddit datasets the reference targets.  On a directed
+        graph the propagation runs against edge direction and the
+        resulting placement can be badly mis-ranked."""
         self.lazy_init_quiver()
         if self.mode == "CPU":
             raise RuntimeError("sample_prob needs a GPU sampler")
