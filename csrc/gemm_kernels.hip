@@ -102,18 +102,24 @@ __device__ __forceinline__ void store_b(float (*Bs)[BP], int tid,
 __global__ void __launch_bounds__(NT)
 tall_gemm_kernel(const float* __restrict__ A, const float* __restrict__ B,
                  const float* __restrict__ bias, float* __restrict__ C,
-                 int64_t M, int K, int N) {
+                 int64_t M, int K, int N, int tn_) {
     __shared__ float As[2][BM][AP];
     __shared__ float Bs[2][BK][BP];
 
-    // m grid-stride: with K <= a few hundred a single 64x64 tile is only
-    // 2-8 k-stages — block setup and pipeline fill dominate (measured
-    // 0.5-0.9x rocBLAS as a one-tile-per-block kernel).  Each block
-    // walks several m-tiles with one CONTINUOUS ping-pong pipeline: the
-    // next tile's first slices prefetch during the current tile's last
-    // k-stage, and the small B panel stays L2-hot across tiles.
-    const int64_t m_stride = (int64_t)gridDim.x * BM;
-    const int n0 = blockIdx.y * BN;
+    // 1-D XCD-aware grid (hardware dispatches block i to XCD i%8): all
+    // tn n-tiles of one m-chunk land on the SAME XCD, so the A slice is
+    // pulled from HBM once and re-read from that XCD's L2 (PMC showed A
+    // re-read x tn was the dominant traffic).  Each block walks several
+    // m-tiles with one CONTINUOUS ping-pong pipeline: the next tile's
+    // first slices prefetch during the current tile's last k-stage.
+    const int n_xcd = 8;
+    const int xcd = blockIdx.x % n_xcd;
+    const int slot = blockIdx.x / n_xcd;
+    const int nt = slot % tn_;
+    const int walker = slot / tn_;          // m-walker within this xcd
+    const int walkers = (int)(gridDim.x / (n_xcd * tn_));
+    const int64_t m_stride = (int64_t)n_xcd * walkers * BM;
+    const int n0 = nt * BN;
     const int tid = threadIdx.x;
     const int wave = tid / 64;
     const int lane = tid % 64;
@@ -126,7 +132,8 @@ tall_gemm_kernel(const float* __restrict__ A, const float* __restrict__ B,
     const float badd = (bias && n_out < N) ? bias[n_out] : 0.f;
 
     float4 ra[2], rb[2];
-    int64_t m0 = (int64_t)blockIdx.x * BM;
+    int64_t m0 = (int64_t)(xcd + n_xcd * walker) * BM;
+    if (m0 >= M) return;
     load_a(A, M, K, m0, 0, tid, ra);
     load_b(B, K, N, 0, n0, tid, rb);
     store_a(As[0], tid, ra);
@@ -183,11 +190,14 @@ void launch_tall_gemm(hipStream_t s, const float* A, const float* B,
     if (M == 0 || K == 0 || N == 0) return;
     int64_t tm = (M + BM - 1) / BM;
     int tn = (N + BN - 1) / BN;
-    // enough blocks to fill 256 CUs several times over, few enough that
-    // each walks multiple m-tiles (pipeline amortization)
-    int gx = (int)std::min<int64_t>(tm, 4096 / tn);
-    dim3 grid((unsigned)gx, (unsigned)tn);
-    tall_gemm_kernel<<<grid, NT, 0, s>>>(A, B, bias, C, M, K, N);
+    // walkers per (xcd, n-tile): enough blocks to fill the chip several
+    // times over, few enough that each walks multiple m-tiles
+    int64_t walkers = 4096 / (8 * tn);
+    int64_t need = (tm + 7) / 8;  // m-tiles per xcd
+    if (walkers > need) walkers = need;
+    if (walkers < 1) walkers = 1;
+    unsigned gx = (unsigned)(8 * tn * walkers);
+    tall_gemm_kernel<<<gx, NT, 0, s>>>(A, B, bias, C, M, K, N, tn);
     QK_CHECK_HIP(hipGetLastError());
 }
 
