@@ -2,9 +2,13 @@
 // a GNN frontier (10^5..10^6 rows) and K, N are layer widths (<= 1024).
 //
 // This is the OTHER hot GEMM family of the model layer (wgrad_kernels.hip
-// covers A^T@B): the GAT projection forward runs x[1.06M×100] @ W^T and
-// its data-grad — rocBLAS picks MT256x64 stream-K tiles and lands ~5.8x
-// off the memory floor (1.12 ms for a read-424MB/write-1.08GB op).
+// covers A^T@B): big-frontier linear forwards and data-grads.  Measured
+// status (benchmarks/bench_gemm.py): WINS on narrow-output shapes
+// (N <= 128: 1.1-1.25x rocBLAS — QLinear routes those here) and loses
+// ~15-45% on wide ones (rocBLAS's 256x128 macro-tiles amortize the
+// barrier convoy better at these shallow K; QLinear keeps rocBLAS
+// there).  PMC: 56% wave-parked / 33% issue-stall — the 64x64 tile's
+// 4-wave barrier per 16-MFMA stage is the pacing structure.
 // MI355X-first design:
 //  - v_mfma_f32_32x32x2_f32 tiles (exact f32), one 32x32 tile per wave,
 //    2x2 waves = a 64x64 C macro-tile per block.  No split-K: M/64 tiles
