@@ -56,6 +56,11 @@ def run(rank, world, sampler, feature, y, train_idx, dim, classes):
 
 def main(world=2, nodes=500_000, edges=10_000_000, dim=100, classes=47):
     import numpy as np
+    avail = torch.cuda.device_count()
+    if world > avail:
+        print(f"# only {avail} GPU(s) visible: clamping world {world} -> "
+              f"{avail}")
+        world = max(1, avail)
     rng = np.random.default_rng(0)
     deg = np.maximum((rng.pareto(1.3, nodes) * 4).astype(np.int64), 1)
     indptr = np.zeros(nodes + 1, dtype=np.int64)

@@ -54,7 +54,7 @@ def main(n=100_000, dim=64, requests=500, batch=64, device_list=None,
                                     sample_mode="Auto", threshold=2000,
                                     neighbour_path=nbr_path)
     hybrid = quiver.HybridSampler(csr_topo, sizes, device_num=nproc,
-                                  worker_num_per_device=2,
+                                  worker_num_per_device=4,
                                   batched_queue_list=
                                   batcher.batched_request_queue_list())
     hybrid.start()
@@ -66,11 +66,19 @@ def main(n=100_000, dim=64, requests=500, batch=64, device_list=None,
     t = threading.Thread(target=server.start, kwargs=dict(join=True))
     t.start()
 
-    time.sleep(5)  # workers warm up
+    # gate offered load on worker warm-up (model load + first CUDA
+    # context) so cold-start doesn't queue up as request backlog
+    ready = server.wait_ready(timeout=120)
+    print(f"{ready}/{server.num_proc} inference workers warm")
+    period = 1.0 / 250  # ~250 QPS offered load
+    t0 = time.perf_counter()
     for i in range(requests):
         ids = rng.integers(0, n, batch)
+        target = t0 + i * period
+        now = time.perf_counter()
+        if target > now:
+            time.sleep(target - now)
         stream_queues[i % nproc].put(ids)
-        time.sleep(0.002)  # ~500 QPS offered load per queue
     time.sleep(3)
     batcher.stop()
     t.join(timeout=120)
