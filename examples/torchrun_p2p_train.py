@@ -15,6 +15,9 @@ owner + cudaIpc reopen (examples/multi_gpu/pyg/ogb-products/
 dist_sampling_ogb_products_quiver.py).
 """
 import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 import torch.distributed as dist
