@@ -69,7 +69,7 @@ def main(n=100_000, dim=64, requests=500, batch=64, device_list=None,
     server = quiver.InferenceServer_Debug(
         model_path, device_list, x, hybrid.sampled_request_queue_list(),
         sample_mode="Auto", csr_topo=csr_topo, sizes=sizes,
-        ignord_length=20, proc_num_per_device=2, uva_gpu="UVA")
+        ignord_length=150, proc_num_per_device=3, uva_gpu="UVA")
     import threading
     t = threading.Thread(target=server.start, kwargs=dict(join=True))
     t.start()
