@@ -435,17 +435,25 @@ class InferenceServer_Debug(InferenceServer):
         lat = arr[:, 2] - arr[:, 0]
         avg_latency = np.average(lat, axis=0, weights=arr[:, 3])
         tp99_latency = np.percentile(lat, 99, axis=0) * 1000
+        # split: queue wait (arrival -> worker pickup) vs service
+        # (pickup -> done) — tells congestion apart from slow work
+        wait = arr[:, 1] - arr[:, 0]
+        serve = arr[:, 2] - arr[:, 1]
         span = max(np.max(arr[:, 2]) - np.min(arr[:, 2]), 1e-9)
         throughput = np.sum(arr[:, 3]) / span
         total = np.sum(arr[:, 3])
         print(f"{tag} Rank {rank}: Avg Latency: {avg_latency}, "
               f"TP99 Latency: {tp99_latency} ms, Throughput: {throughput}, "
-              f"Total: {total}", flush=True)
+              f"Total: {total} "
+              f"(avg wait {np.mean(wait)*1e3:.2f} ms / "
+              f"serve {np.mean(serve)*1e3:.2f} ms)", flush=True)
         if res_path is not None:
             np.save(os.path.join(res_path, f"{exp_id}_{tag}_{rank}"), arr)
         return dict(avg_latency=float(avg_latency),
                     tp99_latency_ms=float(tp99_latency),
-                    throughput=float(throughput), total=float(total))
+                    throughput=float(throughput), total=float(total),
+                    avg_wait_ms=float(np.mean(wait) * 1e3),
+                    avg_serve_ms=float(np.mean(serve) * 1e3))
 
     def gpu_sampler_inference_loop(self, rank, device_list, feature,
                                    gpu_sample_task_queue_list, model_path,
