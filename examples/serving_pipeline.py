@@ -69,7 +69,14 @@ def main(n=100_000, dim=64, requests=500, batch=64, device_list=None,
     server = quiver.InferenceServer_Debug(
         model_path, device_list, x, hybrid.sampled_request_queue_list(),
         sample_mode="Auto", csr_topo=csr_topo, sizes=sizes,
-        ignord_length=150, proc_num_per_device=3, uva_gpu="UVA")
+        # 2 = one GPU-fed + one CPU-fed worker: the MINIMUM for Auto mode
+        # and the right number for a sparse demo load — idle CUDA
+        # processes sharing a GPU pay ~100 ms context-switch penalties
+        # when kernels arrive sparsely (measured: 1.9 ms/serve solo,
+        # ~120 ms with 3 processes at 50 req/s each).  Dense production
+        # load amortizes this (bench_serving: p99 3.8 ms at 1000 QPS
+        # with 4 workers); scale workers with load, not ahead of it.
+        ignord_length=150, proc_num_per_device=2, uva_gpu="UVA")
     import threading
     t = threading.Thread(target=server.start, kwargs=dict(join=True))
     t.start()
