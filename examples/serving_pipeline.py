@@ -21,7 +21,7 @@ import quiver
 from quiver.nn import GraphSAGE
 
 
-def main(n=100_000, dim=64, requests=500, batch=64, device_list=None,
+def main(n=100_000, dim=64, requests=300, batch=64, device_list=None,
          tmpdir="/tmp/quiver_serving"):
     os.makedirs(tmpdir, exist_ok=True)
     use_gpu = torch.cuda.is_available()
@@ -85,7 +85,11 @@ def main(n=100_000, dim=64, requests=500, batch=64, device_list=None,
     # context) so cold-start doesn't queue up as request backlog
     ready = server.wait_ready(timeout=120)
     print(f"{ready}/{server.num_proc} inference workers warm")
-    period = 1.0 / 150  # modest offered load for a demo
+    period = 1.0 / 60  # demo-scale load: with two CUDA contexts
+    # sharing the GPU at SPARSE arrivals, each serve pays ~30 ms of
+    # context switching (profiles/ROUND2_ANALYSIS.md §7) — size the
+    # offered load to the measured capacity, or use dense load +
+    # more workers as benchmarks/bench_serving.py does
     t0 = time.perf_counter()
     for i in range(requests):
         ids = rng.integers(0, n, batch)
