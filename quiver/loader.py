@@ -167,6 +167,16 @@ class TrainingPrefetcher:
                     sides[idx] = torch.cuda.Stream(self.device,
                                                    priority=-1)
             side = sides[idx]
+            # record a main-stream event EVERY produce (graphed or not):
+            # replays wait on the previous produce's event, and a stale
+            # deque after a mix of eager and graphed batches would weaken
+            # that ordering
+            if want_graph:
+                ev_main = torch.cuda.Event()
+                ev_main.record(cur)
+                main_evs.append(ev_main)
+                if len(main_evs) > 2:
+                    main_evs.popleft()
             with torch.cuda.stream(side):
                 if chain_async[0] and isinstance(graphed[0], _GraphedChain) \
                         and seeds.numel() == graphed[0].batch_size:
@@ -177,11 +187,6 @@ class TrainingPrefetcher:
                     # previous produce's main-stream event (not this
                     # one's) keeps the replay overlapping the current
                     # batch's compute.
-                    ev_main = torch.cuda.Event()
-                    ev_main.record(cur)
-                    main_evs.append(ev_main)
-                    if len(main_evs) > 2:
-                        main_evs.popleft()
                     side.wait_event(main_evs[0])
                     try:
                         tok, x_ub = graphed[0].run(seeds)
