@@ -239,13 +239,16 @@ class SampleJob(Generic[T_co]):
 
 
 def cpu_sampler_worker_loop(rank, quiver_sampler, task_queue, result_queue):
+    import torch as _torch
+    _torch.set_num_threads(1)  # pool parallelism comes from worker count
     while True:
         task = task_queue.get()
         if isinstance(task, _StopWork):
             result_queue.put(_StopWork())
             break
+        t0 = time.time()
         res = quiver_sampler.sample(task)
-        result_queue.put(res)
+        result_queue.put((res, time.time() - t0))
 
 
 class MixedGraphSageSampler:
@@ -295,13 +298,20 @@ class MixedGraphSageSampler:
                 self.workers.append(p)
         self.inited = True
 
-    def decide_task_num(self, dev_time, dev_tasks, cpu_time, cpu_tasks):
-        # adaptive split: how many tasks CPU workers get per GPU task batch
-        if cpu_time <= 0 or cpu_tasks == 0:
+    def decide_task_num(self, dev_time, dev_tasks, cpu_service, cpu_tasks):
+        """How many tasks to hand the CPU pool per GPU inline task.
+
+        Measured-time split (reference sage_sampler.py:272-288): the pool
+        can finish num_workers * t_dev / t_cpu tasks while the device
+        samples one; half that keeps the tail from ending on a slow CPU
+        task.  Until both sides have measurements, seed every worker one
+        task so the first estimate exists.
+        """
+        if cpu_tasks == 0 or dev_tasks == 0:
             return max(1, self.num_workers)
-        dev_rate = dev_tasks / max(dev_time, 1e-9)
-        cpu_rate = cpu_tasks / max(cpu_time, 1e-9)
-        return max(0, int(cpu_rate / max(dev_rate, 1e-9) * dev_tasks / 2))
+        t_dev = dev_time / dev_tasks          # device seconds per task
+        t_cpu = cpu_service / cpu_tasks       # ONE worker's seconds per task
+        return max(0, int(self.num_workers * t_dev / max(t_cpu, 1e-9) / 2))
 
     def __iter__(self):
         self.lazy_init()
@@ -312,23 +322,26 @@ class MixedGraphSageSampler:
         n = len(self.sample_job)
         next_task = 0
         pending_cpu = 0
-        dev_time, dev_done = 1e-9, 0
-        cpu_time, cpu_done = 0.0, 0
-        cpu_t0 = None
+        dev_time, dev_done = 0.0, 0
+        cpu_service, cpu_done = 0.0, 0   # summed per-task worker seconds
+        rr = 0                           # round-robin worker cursor
         try:
             while next_task < n or pending_cpu > 0:
-                # hand a slice to CPU workers
+                # hand a slice to CPU workers (round-robin so the pool
+                # load-balances instead of piling on worker 0)
                 if self.task_queues and next_task < n:
-                    want = self.decide_task_num(dev_time, max(dev_done, 1),
-                                                cpu_time, cpu_done)
-                    for w in range(min(want, n - next_task,
-                                       len(self.task_queues))):
-                        self.task_queues[w].put(
+                    want = self.decide_task_num(dev_time, dev_done,
+                                                cpu_service, cpu_done)
+                    # cap queue depth: a task handed out now is committed
+                    # even if the measured rates later say otherwise
+                    want = min(want, n - next_task,
+                               2 * len(self.task_queues) - pending_cpu)
+                    for _ in range(max(0, want)):
+                        self.task_queues[rr % len(self.task_queues)].put(
                             self.sample_job[next_task])
+                        rr += 1
                         next_task += 1
                         pending_cpu += 1
-                    if cpu_t0 is None and pending_cpu:
-                        cpu_t0 = time.time()
                 # GPU samples inline
                 if next_task < n:
                     t0 = time.time()
@@ -347,10 +360,12 @@ class MixedGraphSageSampler:
                         break
                     if isinstance(res, _StopWork):
                         continue
+                    if isinstance(res, tuple) and len(res) == 2 \
+                            and isinstance(res[1], float):
+                        res, dt = res
+                        cpu_service += dt
                     pending_cpu -= 1
                     cpu_done += 1
-                    if cpu_t0 is not None:
-                        cpu_time = time.time() - cpu_t0
                     yield res
                     if next_task < n:
                         break
