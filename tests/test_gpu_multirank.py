@@ -116,6 +116,16 @@ def _dist_feature_worker(rank, world, q, rdv_file):
             if not torch.equal(got, feat[ids]):
                 bad = (got != feat[ids]).any(1).sum().item()
                 raise AssertionError(f"{bad}/257 gathered rows wrong")
+        # the async-chain path (upper-bound ids + device-side exact count)
+        # across the hipIpc-assembled shards — what TrainingPrefetcher's
+        # zero-sync chain runs at N>1
+        ids = torch.randint(0, n, (200,), generator=g)
+        ub = torch.zeros(256, dtype=torch.long)
+        ub[:200] = ids
+        n_dev_t = torch.tensor([200], dtype=torch.long, device=dev)
+        x_ub = f.gather_raw(ub.to(dev), n_dev_t)
+        if not torch.equal(x_ub[:200].cpu(), feat[ids]):
+            raise AssertionError("gather_raw rows wrong across IPC shards")
         dist.barrier()  # peers may still be reading our shard
         q.put((rank, "ok"))
         dist.destroy_process_group()
