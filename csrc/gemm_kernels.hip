@@ -294,7 +294,11 @@ tall_gemm128_kernel(const float* __restrict__ A, const float* __restrict__ B,
 void launch_tall_gemm(hipStream_t s, const float* A, const float* B,
                       const float* bias, float* C, int64_t M, int K, int N) {
     if (M == 0 || K == 0 || N == 0) return;
-    const bool big = M >= 262144;  // huge-frontier shapes: 128x128 tiles
+    // 128x128 tiles measured SLOWER on the huge-frontier shapes (1461
+    // vs 1111 us at 1.06Mx100x256): LDS doubles -> 2 blocks/CU and the
+    // lost concurrency outweighs 4x compute per barrier.  Kernel kept
+    // for reference/perf archaeology; routing stays on 64x64.
+    const bool big = false;
     const int bm = big ? BM2 : BM, bn = big ? BN2 : BN;
     int64_t tm = (M + bm - 1) / bm;
     int tn = (N + bn - 1) / bn;
