@@ -308,6 +308,14 @@ class GATConv(nn.Module):
         n_dst = x_dst.size(0) if size is None else int(size[1])
 
         h2 = self.lin(x_src)
+        # bf16 models: projection GEMM and the feature path stay bf16 (the
+        # bandwidth win); attention math runs through the fp32 fused
+        # kernels via one boundary cast — torch's bf16 fallback chain
+        # (scatter_reduce amax -> ~120 rocprim sort kernels per step) is
+        # far slower than the cast
+        out_dtype = h2.dtype
+        if h2.is_cuda and h2.dtype == torch.bfloat16 and self.sorted_dst:
+            h2 = h2.float()
         h_src = h2.view(-1, H, C)
         prefix = (x_dst.data_ptr() == x_src.data_ptr()
                   and x_dst.size(0) <= x_src.size(0)
@@ -318,8 +326,8 @@ class GATConv(nn.Module):
             # bipartite prefix convention (x_dst is x_src[:n_dst]) with
             # the fused dots kernel: one read of h for both logit sets
             alpha_src, alpha_dst = _GatDots.apply(
-                h2, self.att_src.reshape(-1), self.att_dst.reshape(-1),
-                x_dst.size(0), H)
+                h2, self.att_src.reshape(-1).float(),
+                self.att_dst.reshape(-1).float(), x_dst.size(0), H)
         else:
             if prefix:
                 h_dst = h_src[:x_dst.size(0)]
@@ -368,7 +376,7 @@ class GATConv(nn.Module):
                    else out.mean(1))
         if self.bias is not None:
             out = out + self.bias
-        return out
+        return out.to(out_dtype) if out.dtype != out_dtype else out
 
 
 class GraphSAGE(nn.Module):

@@ -252,3 +252,34 @@ def test_gat_dots_matches_torch():
     assert torch.allclose(h.grad, h2.grad, atol=1e-4, rtol=1e-4)
     assert torch.allclose(a_s.grad, a_s2.grad, atol=2e-3, rtol=1e-3)
     assert torch.allclose(a_d.grad, a_d2.grad, atol=2e-3, rtol=1e-3)
+
+
+def test_gat_bf16_e2e_step():
+    """bf16 GAT: projection/gather bf16, attention via the fp32 fused
+    kernels behind a boundary cast — one fwd+bwd step must be finite and
+    outputs bf16."""
+    from quiver.nn import GAT
+    import quiver
+    g = torch.Generator().manual_seed(0)
+    n = 3000
+    deg = torch.randint(1, 20, (n,), generator=g)
+    indptr = torch.zeros(n + 1, dtype=torch.long)
+    torch.cumsum(deg, 0, out=indptr[1:])
+    indices = torch.randint(0, n, (int(indptr[-1]),), generator=g)
+    topo = quiver.CSRTopo(indptr=indptr, indices=indices)
+    s = quiver.GraphSageSampler(topo, [8, 4], device=0, mode="GPU")
+    n_id, bs, adjs = s.sample(torch.arange(64))
+    x = torch.randn(n_id.numel(), 100, generator=g) \
+        .to(torch.bfloat16).cuda().requires_grad_(True)
+    model = GAT(100, 64, 10, num_layers=2, heads=4, dropout=0.0) \
+        .to(torch.bfloat16).cuda()
+    out = model(x, adjs)
+    assert out.dtype == torch.bfloat16
+    loss = torch.nn.functional.nll_loss(
+        out.float(), torch.randint(0, 10, (bs,), device="cuda"))
+    loss.backward()
+    assert torch.isfinite(x.grad.float()).all()
+    for name, p in model.named_parameters():
+        if p.grad is not None:
+            assert p.grad.dtype == p.dtype, name
+            assert torch.isfinite(p.grad.float()).all(), name
