@@ -1203,7 +1203,9 @@ std::tuple<torch::Tensor, torch::Tensor> wgrad(torch::Tensor a,
 torch::Tensor segment_wsum(torch::Tensor x, torch::Tensor w,
                            torch::Tensor src, torch::Tensor dst_ptr,
                            int64_t heads) {
-    TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kFloat32 && x.dim() == 2);
+    TORCH_CHECK(x.is_cuda() && x.dim() == 2 &&
+                (x.dtype() == torch::kFloat32 ||
+                 x.dtype() == torch::kBFloat16));
     TORCH_CHECK(w.is_cuda() && w.dtype() == torch::kFloat32 && w.dim() == 2 &&
                 w.size(1) == heads);
     x = x.contiguous();
@@ -1214,6 +1216,13 @@ torch::Tensor segment_wsum(torch::Tensor x, torch::Tensor w,
     int chead = (int)(x.size(1) / heads);
     TORCH_CHECK((int64_t)chead * heads == x.size(1), "dim % heads != 0");
     auto out = torch::empty({n_dst, x.size(1)}, x.options());
+    if (x.dtype() == torch::kBFloat16) {
+        qk::launch_segment_wsum_fwd_bf16(
+            current_stream(), x.data_ptr(), w.data_ptr<float>(),
+            src.data_ptr<int64_t>(), dst_ptr.data_ptr<int64_t>(), n_dst,
+            (int)heads, chead, out.data_ptr());
+        return out;
+    }
     qk::launch_segment_wsum_fwd(current_stream(), x.data_ptr<float>(),
                                 w.data_ptr<float>(), src.data_ptr<int64_t>(),
                                 dst_ptr.data_ptr<int64_t>(), n_dst,
@@ -1232,22 +1241,39 @@ std::tuple<torch::Tensor, torch::Tensor> segment_wsum_backward(
     dst_ptr = dst_ptr.contiguous();
     int64_t n_dst = dst_ptr.numel() - 1;
     int chead = (int)(x.size(1) / heads);
+    const bool bf16 = x.dtype() == torch::kBFloat16;
     torch::Tensor gx, gw;
     if (need_gx) {
         gx = torch::zeros_like(x);
-        qk::launch_segment_wsum_bwd_x(
-            current_stream(), grad_out.data_ptr<float>(),
-            w.data_ptr<float>(), src.data_ptr<int64_t>(),
-            dst_ptr.data_ptr<int64_t>(), n_dst, (int)heads, chead,
-            gx.data_ptr<float>());
+        if (bf16) {
+            TORCH_CHECK(((int64_t)chead * heads) % 2 == 0,
+                        "bf16 wsum needs an even dim");
+            qk::launch_segment_wsum_bwd_x_bf16(
+                current_stream(), grad_out.data_ptr(), w.data_ptr<float>(),
+                src.data_ptr<int64_t>(), dst_ptr.data_ptr<int64_t>(), n_dst,
+                (int)heads, chead, gx.data_ptr());
+        } else {
+            qk::launch_segment_wsum_bwd_x(
+                current_stream(), grad_out.data_ptr<float>(),
+                w.data_ptr<float>(), src.data_ptr<int64_t>(),
+                dst_ptr.data_ptr<int64_t>(), n_dst, (int)heads, chead,
+                gx.data_ptr<float>());
+        }
     }
     if (need_gw) {
         gw = torch::empty_like(w);
-        qk::launch_segment_wsum_bwd_w(
-            current_stream(), grad_out.data_ptr<float>(),
-            x.data_ptr<float>(), src.data_ptr<int64_t>(),
-            dst_ptr.data_ptr<int64_t>(), n_dst, (int)heads, chead,
-            gw.data_ptr<float>());
+        if (bf16) {
+            qk::launch_segment_wsum_bwd_w_bf16(
+                current_stream(), grad_out.data_ptr(), x.data_ptr(),
+                src.data_ptr<int64_t>(), dst_ptr.data_ptr<int64_t>(), n_dst,
+                (int)heads, chead, gw.data_ptr<float>());
+        } else {
+            qk::launch_segment_wsum_bwd_w(
+                current_stream(), grad_out.data_ptr<float>(),
+                x.data_ptr<float>(), src.data_ptr<int64_t>(),
+                dst_ptr.data_ptr<int64_t>(), n_dst, (int)heads, chead,
+                gw.data_ptr<float>());
+        }
     }
     return {gx, gw};
 }
@@ -1362,7 +1388,12 @@ std::tuple<torch::Tensor, torch::Tensor> gat_dots(torch::Tensor h,
                                                   torch::Tensor att_dst,
                                                   int64_t n_dst,
                                                   int64_t heads) {
-    TORCH_CHECK(h.is_cuda() && h.dtype() == torch::kFloat32 && h.dim() == 2);
+    TORCH_CHECK(h.is_cuda() && h.dim() == 2 &&
+                (h.dtype() == torch::kFloat32 ||
+                 h.dtype() == torch::kBFloat16));
+    TORCH_CHECK(att_src.dtype() == torch::kFloat32 &&
+                att_dst.dtype() == torch::kFloat32,
+                "attention vectors stay fp32 (logits are fp32)");
     h = h.contiguous();
     att_src = att_src.contiguous();
     att_dst = att_dst.contiguous();
@@ -1371,9 +1402,11 @@ std::tuple<torch::Tensor, torch::Tensor> gat_dots(torch::Tensor h,
     TORCH_CHECK((int64_t)chead * heads == h.size(1), "dim % heads != 0");
     TORCH_CHECK(att_src.numel() == h.size(1) &&
                 att_dst.numel() == h.size(1));
-    auto asrc = torch::empty({n, heads}, h.options());
-    auto adst = torch::empty({n_dst, heads}, h.options());
-    qk::launch_gat_dots_fwd(current_stream(), h.data_ptr<float>(),
+    auto fopts = h.options().dtype(torch::kFloat32);
+    auto asrc = torch::empty({n, heads}, fopts);
+    auto adst = torch::empty({n_dst, heads}, fopts);
+    qk::launch_gat_dots_fwd(current_stream(), h.data_ptr(),
+                            h.dtype() == torch::kBFloat16,
                             att_src.data_ptr<float>(),
                             att_dst.data_ptr<float>(), n, n_dst, (int)heads,
                             chead, asrc.data_ptr<float>(),
@@ -1395,10 +1428,10 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> gat_dots_backward(
     auto g_as = torch::zeros_like(att_src);
     auto g_ad = torch::zeros_like(att_dst);
     qk::launch_gat_dots_bwd(
-        current_stream(), h.data_ptr<float>(), att_src.data_ptr<float>(),
-        att_dst.data_ptr<float>(), g_asrc.data_ptr<float>(),
-        g_adst.data_ptr<float>(), n, n_dst, (int)heads, chead,
-        g_h.data_ptr<float>(), g_as.data_ptr<float>(),
+        current_stream(), h.data_ptr(), h.dtype() == torch::kBFloat16,
+        att_src.data_ptr<float>(), att_dst.data_ptr<float>(),
+        g_asrc.data_ptr<float>(), g_adst.data_ptr<float>(), n, n_dst,
+        (int)heads, chead, g_h.data_ptr(), g_as.data_ptr<float>(),
         g_ad.data_ptr<float>());
     return {g_h, g_as, g_ad};
 }

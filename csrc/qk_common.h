@@ -141,15 +141,16 @@ void launch_scatter(hipStream_t s, const GatherSpec& spec,
 // Fused GAT attention dots: asrc[n,h] = <h[n,h,:], att_src[h,:]>, adst
 // over the first n_dst rows (prefix convention); backward also emits
 // g_h and the (tiny) att gradients via per-block LDS accumulation.
-void launch_gat_dots_fwd(hipStream_t s, const float* h,
+// h (and g_h) may be fp32 or bf16 (h_bf16); logits/att stay fp32.
+void launch_gat_dots_fwd(hipStream_t s, const void* h, bool h_bf16,
                          const float* att_src, const float* att_dst,
                          int64_t n, int64_t n_dst, int heads, int chead,
                          float* asrc, float* adst);
-void launch_gat_dots_bwd(hipStream_t s, const float* h,
+void launch_gat_dots_bwd(hipStream_t s, const void* h, bool h_bf16,
                          const float* att_src, const float* att_dst,
                          const float* g_asrc, const float* g_adst,
                          int64_t n, int64_t n_dst, int heads, int chead,
-                         float* g_h, float* g_att_src, float* g_att_dst);
+                         void* g_h, float* g_att_src, float* g_att_dst);
 
 // ---------- tall-M GEMM (gemm_kernels.hip) -------------------------------
 
@@ -195,6 +196,18 @@ void launch_segment_mean_bwd(hipStream_t s, const float* grad_out,
 // Weighted segment sum (GAT attention aggregation), x [n_src, H*C],
 // w [n_edges, H] (per-edge per-head attention):
 //   out[d, h*C+c] = sum_{e in segment(d)} w[e,h] * x[src[e], h*C+c]
+void launch_segment_wsum_fwd_bf16(hipStream_t s, const void* x,
+                                  const float* w, const int64_t* src,
+                                  const int64_t* dst_ptr, int64_t n_dst,
+                                  int heads, int chead, void* out);
+void launch_segment_wsum_bwd_x_bf16(hipStream_t s, const void* grad_out,
+                                    const float* w, const int64_t* src,
+                                    const int64_t* dst_ptr, int64_t n_dst,
+                                    int heads, int chead, void* grad_x);
+void launch_segment_wsum_bwd_w_bf16(hipStream_t s, const void* grad_out,
+                                    const void* x, const int64_t* src,
+                                    const int64_t* dst_ptr, int64_t n_dst,
+                                    int heads, int chead, float* grad_w);
 void launch_segment_wsum_fwd(hipStream_t s, const float* x, const float* w,
                              const int64_t* src, const int64_t* dst_ptr,
                              int64_t n_dst, int heads, int chead, float* out);

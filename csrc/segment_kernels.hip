@@ -23,6 +23,39 @@ constexpr int SUB = 16;                     // lanes per dst segment
 constexpr int ROWS_PER_BLOCK = BLOCK / SUB;
 constexpr int VPL = 4;                      // floats per lane per chunk
 
+// 4-element load/store in fp32 working precision for fp32 and bf16
+// element types (bf16 goes through 4-byte bf16x2 pairs: aligned for any
+// even dim; the value math stays fp32 and only stores round).
+template <typename T> struct vec4io;
+template <> struct vec4io<float> {
+    static __device__ __forceinline__ void load(const float* p, float v[4]) {
+        const float4 t = *reinterpret_cast<const float4*>(p);
+        v[0] = t.x; v[1] = t.y; v[2] = t.z; v[3] = t.w;
+    }
+    static __device__ __forceinline__ void store(float* p, const float v[4]) {
+        *reinterpret_cast<float4*>(p) = float4{v[0], v[1], v[2], v[3]};
+    }
+};
+template <> struct vec4io<__hip_bfloat16> {
+    static __device__ __forceinline__ void load(const __hip_bfloat16* p,
+                                                float v[4]) {
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            const float2 f = __bfloat1622float2(
+                *reinterpret_cast<const __hip_bfloat162*>(p + 2 * i));
+            v[2 * i] = f.x;
+            v[2 * i + 1] = f.y;
+        }
+    }
+    static __device__ __forceinline__ void store(__hip_bfloat16* p,
+                                                 const float v[4]) {
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+            *reinterpret_cast<__hip_bfloat162*>(p + 2 * i) =
+                __float22bfloat162_rn(float2{v[2 * i], v[2 * i + 1]});
+    }
+};
+
 __global__ void __launch_bounds__(BLOCK)
 segment_mean_fwd_kernel(const float* __restrict__ x,
                         const int64_t* __restrict__ src,
@@ -179,12 +212,13 @@ segment_mean_bwd_bf16_kernel(const __hip_bfloat16* __restrict__ grad_out,
 // VPL in the models here, so a VPL chunk never straddles heads — guarded
 // by the launcher).
 
+template <typename XT>
 __global__ void __launch_bounds__(BLOCK)
-segment_wsum_fwd_kernel(const float* __restrict__ x,
+segment_wsum_fwd_kernel(const XT* __restrict__ x,
                         const float* __restrict__ w,
                         const int64_t* __restrict__ src,
                         const int64_t* __restrict__ dst_ptr, int64_t n_dst,
-                        int heads, int chead, float* __restrict__ out) {
+                        int heads, int chead, XT* __restrict__ out) {
     const int64_t dim = (int64_t)heads * chead;
     const int sub_id = threadIdx.x / SUB;
     const int lane = threadIdx.x % SUB;
@@ -198,21 +232,22 @@ segment_wsum_fwd_kernel(const float* __restrict__ x,
             const int wd = (int)min((int64_t)VPL, dim - c);
             for (int64_t e = beg; e < end; ++e) {
                 const float we = w[e * heads + h];
-                const float* row = x + src[e] * dim + c;
+                const XT* row = x + src[e] * dim + c;
                 if (wd == VPL) {
-                    const float4 v = *reinterpret_cast<const float4*>(row);
-                    acc[0] += we * v.x; acc[1] += we * v.y;
-                    acc[2] += we * v.z; acc[3] += we * v.w;
+                    float v[VPL];
+                    vec4io<XT>::load(row, v);
+#pragma unroll
+                    for (int q = 0; q < VPL; ++q) acc[q] += we * v[q];
                 } else {
-                    for (int q = 0; q < wd; ++q) acc[q] += we * row[q];
+                    for (int q = 0; q < wd; ++q)
+                        acc[q] += we * (float)row[q];
                 }
             }
-            float* orow = out + d * dim + c;
+            XT* orow = out + d * dim + c;
             if (wd == VPL) {
-                float4 v{acc[0], acc[1], acc[2], acc[3]};
-                *reinterpret_cast<float4*>(orow) = v;
+                vec4io<XT>::store(orow, acc);
             } else {
-                for (int q = 0; q < wd; ++q) orow[q] = acc[q];
+                for (int q = 0; q < wd; ++q) orow[q] = (XT)acc[q];
             }
         }
     }
@@ -241,9 +276,44 @@ segment_wsum_bwd_x_kernel(const float* __restrict__ grad_out,
     }
 }
 
+// bf16 variant: grad_out and grad_x are bf16; weights fp32.  Lanes own
+// CHANNEL PAIRS so the scatter uses the packed-bf16 global atomic
+// (dim % 2 == 0 guarded by the launcher; chead % 2 == 0 keeps a pair
+// within one head when heads > 1 -- implied by the chead % 4 guard).
 __global__ void __launch_bounds__(BLOCK)
-segment_wsum_bwd_w_kernel(const float* __restrict__ grad_out,
-                          const float* __restrict__ x,
+segment_wsum_bwd_x_bf16_kernel(const __hip_bfloat16* __restrict__ grad_out,
+                               const float* __restrict__ w,
+                               const int64_t* __restrict__ src,
+                               const int64_t* __restrict__ dst_ptr,
+                               int64_t n_dst, int heads, int chead,
+                               __hip_bfloat16* __restrict__ grad_x) {
+    const int64_t dim = (int64_t)heads * chead;
+    const int sub_id = threadIdx.x / SUB;
+    const int lane = threadIdx.x % SUB;
+    int64_t d = (int64_t)blockIdx.x * ROWS_PER_BLOCK + sub_id;
+    const int64_t stride = (int64_t)gridDim.x * ROWS_PER_BLOCK;
+    for (; d < n_dst; d += stride) {
+        const int64_t beg = dst_ptr[d], end = dst_ptr[d + 1];
+        const __hip_bfloat16* orow = grad_out + d * dim;
+        for (int64_t e = beg; e < end; ++e) {
+            __hip_bfloat16* grow = grad_x + src[e] * dim;
+            const float* wrow = w + e * heads;
+            for (int64_t c = 2 * lane; c + 1 < dim; c += 2 * SUB) {
+                const float we = wrow[c / chead];
+                const float2 g = __bfloat1622float2(
+                    *reinterpret_cast<const __hip_bfloat162*>(&orow[c]));
+                unsafeAtomicAdd(
+                    reinterpret_cast<__hip_bfloat162*>(&grow[c]),
+                    __float22bfloat162_rn(float2{we * g.x, we * g.y}));
+            }
+        }
+    }
+}
+
+template <typename XT>
+__global__ void __launch_bounds__(BLOCK)
+segment_wsum_bwd_w_kernel(const XT* __restrict__ grad_out,
+                          const XT* __restrict__ x,
                           const int64_t* __restrict__ src,
                           const int64_t* __restrict__ dst_ptr, int64_t n_dst,
                           int heads, int chead, float* __restrict__ grad_w) {
@@ -255,14 +325,14 @@ segment_wsum_bwd_w_kernel(const float* __restrict__ grad_out,
     const int64_t stride = (int64_t)gridDim.x * ROWS_PER_BLOCK;
     for (; d < n_dst; d += stride) {
         const int64_t beg = dst_ptr[d], end = dst_ptr[d + 1];
-        const float* orow = grad_out + d * dim;
+        const XT* orow = grad_out + d * dim;
         for (int64_t e = beg; e < end; ++e) {
-            const float* xrow = x + src[e] * dim;
+            const XT* xrow = x + src[e] * dim;
             for (int h = 0; h < heads; ++h) {
                 float acc = 0.f;
                 const int64_t base = (int64_t)h * chead;
                 for (int c = lane; c < chead; c += SUB)
-                    acc += orow[base + c] * xrow[base + c];
+                    acc += (float)orow[base + c] * (float)xrow[base + c];
                 for (int off = SUB / 2; off > 0; off >>= 1)
                     acc += __shfl_down(acc, off, SUB);
                 if (lane == 0) grad_w[e * heads + h] = acc;
@@ -399,8 +469,9 @@ gat_alpha_bwd_kernel(const float* __restrict__ grad_alpha,
 // write + 1 GB reduce read per term) with ONE read of h.
 constexpr int DSUB = 16;  // lanes per (row, head) pair
 
+template <typename HT>
 __global__ void __launch_bounds__(BLOCK)
-gat_dots_fwd_kernel(const float* __restrict__ h,
+gat_dots_fwd_kernel(const HT* __restrict__ h,
                     const float* __restrict__ att_src,
                     const float* __restrict__ att_dst, int64_t n,
                     int64_t n_dst, int heads, int chead,
@@ -414,18 +485,19 @@ gat_dots_fwd_kernel(const float* __restrict__ h,
     for (; p < total; p += stride) {
         const int64_t row = p / heads;
         const int hd = (int)(p % heads);
-        const float* hrow = h + (row * heads + hd) * (int64_t)chead;
+        const HT* hrow = h + (row * heads + hd) * (int64_t)chead;
         const float* as = att_src + hd * chead;
         const float* ad = att_dst + hd * chead;
         float sv = 0.f, dv = 0.f;
         const bool do_dst = row < n_dst;
         for (int c = lane * 4; c < chead; c += DSUB * 4) {
-            const float4 v = *reinterpret_cast<const float4*>(hrow + c);
+            float v[4];
+            vec4io<HT>::load(hrow + c, v);
             const float4 a = *reinterpret_cast<const float4*>(as + c);
-            sv += v.x * a.x + v.y * a.y + v.z * a.z + v.w * a.w;
+            sv += v[0] * a.x + v[1] * a.y + v[2] * a.z + v[3] * a.w;
             if (do_dst) {
                 const float4 b = *reinterpret_cast<const float4*>(ad + c);
-                dv += v.x * b.x + v.y * b.y + v.z * b.z + v.w * b.w;
+                dv += v[0] * b.x + v[1] * b.y + v[2] * b.z + v[3] * b.w;
             }
         }
         for (int off = DSUB / 2; off; off >>= 1) {
@@ -442,14 +514,15 @@ gat_dots_fwd_kernel(const float* __restrict__ h,
 // g_h[row,h,:] = g_asrc[row,h]*att_src[h,:] (+ g_adst part for prefix
 // rows); g_att_* accumulate per-block in LDS, one global atomic per
 // element per block (the [H*C] output is tiny next to the 1M-row input).
+template <typename HT>
 __global__ void __launch_bounds__(BLOCK)
-gat_dots_bwd_kernel(const float* __restrict__ h,
+gat_dots_bwd_kernel(const HT* __restrict__ h,
                     const float* __restrict__ att_src,
                     const float* __restrict__ att_dst,
                     const float* __restrict__ g_asrc,
                     const float* __restrict__ g_adst, int64_t n,
                     int64_t n_dst, int heads, int chead,
-                    float* __restrict__ g_h, float* __restrict__ g_att_src,
+                    HT* __restrict__ g_h, float* __restrict__ g_att_src,
                     float* __restrict__ g_att_dst) {
     extern __shared__ float lacc[];  // [2][heads*chead]
     const int D = heads * chead;
@@ -466,34 +539,29 @@ gat_dots_bwd_kernel(const float* __restrict__ h,
     for (; p < total; p += stride) {
         const int64_t row = p / heads;
         const int hd = (int)(p % heads);
-        const float* hrow = h + (row * heads + hd) * (int64_t)chead;
-        float* grow = g_h + (row * heads + hd) * (int64_t)chead;
+        const HT* hrow = h + (row * heads + hd) * (int64_t)chead;
+        HT* grow = g_h + (row * heads + hd) * (int64_t)chead;
         const float* as = att_src + hd * chead;
         const float* ad = att_dst + hd * chead;
         const bool do_dst = row < n_dst;
         const float gs = g_asrc[row * heads + hd];
         const float gd = do_dst ? g_adst[row * heads + hd] : 0.f;
         for (int c = lane * 4; c < chead; c += DSUB * 4) {
-            const float4 v = *reinterpret_cast<const float4*>(hrow + c);
+            float v[4];
+            vec4io<HT>::load(hrow + c, v);
             const float4 a = *reinterpret_cast<const float4*>(as + c);
-            float4 g;
-            g.x = gs * a.x; g.y = gs * a.y; g.z = gs * a.z; g.w = gs * a.w;
+            float g[4] = {gs * a.x, gs * a.y, gs * a.z, gs * a.w};
             if (do_dst) {
                 const float4 b = *reinterpret_cast<const float4*>(ad + c);
-                g.x += gd * b.x; g.y += gd * b.y;
-                g.z += gd * b.z; g.w += gd * b.w;
+                g[0] += gd * b.x; g[1] += gd * b.y;
+                g[2] += gd * b.z; g[3] += gd * b.w;
             }
-            *reinterpret_cast<float4*>(grow + c) = g;
+            vec4io<HT>::store(grow + c, g);
             const int base = hd * chead + c;
-            atomicAdd(&lsrc[base + 0], gs * v.x);
-            atomicAdd(&lsrc[base + 1], gs * v.y);
-            atomicAdd(&lsrc[base + 2], gs * v.z);
-            atomicAdd(&lsrc[base + 3], gs * v.w);
-            if (do_dst) {
-                atomicAdd(&ldst[base + 0], gd * v.x);
-                atomicAdd(&ldst[base + 1], gd * v.y);
-                atomicAdd(&ldst[base + 2], gd * v.z);
-                atomicAdd(&ldst[base + 3], gd * v.w);
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                atomicAdd(&lsrc[base + q], gs * v[q]);
+                if (do_dst) atomicAdd(&ldst[base + q], gd * v[q]);
             }
         }
     }
@@ -567,8 +635,22 @@ void launch_segment_wsum_fwd(hipStream_t s, const float* x, const float* w,
                              int64_t n_dst, int heads, int chead, float* out) {
     if (n_dst == 0) return;
     check_chead(heads, chead);
-    segment_wsum_fwd_kernel<<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0, s>>>(
-        x, w, src, dst_ptr, n_dst, heads, chead, out);
+    segment_wsum_fwd_kernel<float>
+        <<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0, s>>>(
+            x, w, src, dst_ptr, n_dst, heads, chead, out);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+void launch_segment_wsum_fwd_bf16(hipStream_t s, const void* x,
+                                  const float* w, const int64_t* src,
+                                  const int64_t* dst_ptr, int64_t n_dst,
+                                  int heads, int chead, void* out) {
+    if (n_dst == 0) return;
+    check_chead(heads, chead);
+    segment_wsum_fwd_kernel<__hip_bfloat16>
+        <<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0, s>>>(
+            (const __hip_bfloat16*)x, w, src, dst_ptr, n_dst, heads, chead,
+            (__hip_bfloat16*)out);
     QK_CHECK_HIP(hipGetLastError());
 }
 
@@ -584,15 +666,41 @@ void launch_segment_wsum_bwd_x(hipStream_t s, const float* grad_out,
     QK_CHECK_HIP(hipGetLastError());
 }
 
+void launch_segment_wsum_bwd_x_bf16(hipStream_t s, const void* grad_out,
+                                    const float* w, const int64_t* src,
+                                    const int64_t* dst_ptr, int64_t n_dst,
+                                    int heads, int chead, void* grad_x) {
+    if (n_dst == 0) return;
+    check_chead(heads, chead);
+    segment_wsum_bwd_x_bf16_kernel
+        <<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0, s>>>(
+            (const __hip_bfloat16*)grad_out, w, src, dst_ptr, n_dst, heads,
+            chead, (__hip_bfloat16*)grad_x);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
 void launch_segment_wsum_bwd_w(hipStream_t s, const float* grad_out,
                                const float* x, const int64_t* src,
                                const int64_t* dst_ptr, int64_t n_dst,
                                int heads, int chead, float* grad_w) {
     if (n_dst == 0) return;
     check_chead(heads, chead);
-    segment_wsum_bwd_w_kernel<<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0,
-                                s>>>(grad_out, x, src, dst_ptr, n_dst, heads,
-                                     chead, grad_w);
+    segment_wsum_bwd_w_kernel<float>
+        <<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0, s>>>(
+            grad_out, x, src, dst_ptr, n_dst, heads, chead, grad_w);
+    QK_CHECK_HIP(hipGetLastError());
+}
+
+void launch_segment_wsum_bwd_w_bf16(hipStream_t s, const void* grad_out,
+                                    const void* x, const int64_t* src,
+                                    const int64_t* dst_ptr, int64_t n_dst,
+                                    int heads, int chead, float* grad_w) {
+    if (n_dst == 0) return;
+    check_chead(heads, chead);
+    segment_wsum_bwd_w_kernel<__hip_bfloat16>
+        <<<grid_for(n_dst, ROWS_PER_BLOCK), BLOCK, 0, s>>>(
+            (const __hip_bfloat16*)grad_out, (const __hip_bfloat16*)x, src,
+            dst_ptr, n_dst, heads, chead, grad_w);
     QK_CHECK_HIP(hipGetLastError());
 }
 
@@ -638,33 +746,48 @@ void launch_gat_alpha_bwd(hipStream_t s, const float* grad_alpha,
 }
 
 
-void launch_gat_dots_fwd(hipStream_t s, const float* h,
+void launch_gat_dots_fwd(hipStream_t s, const void* h, bool h_bf16,
                          const float* att_src, const float* att_dst,
                          int64_t n, int64_t n_dst, int heads, int chead,
                          float* asrc, float* adst) {
     if (n == 0) return;
     if (chead % 4 != 0)
         throw std::runtime_error("gat_dots: chead must be a multiple of 4");
-    gat_dots_fwd_kernel<<<grid_for(n * heads, BLOCK / DSUB), BLOCK, 0, s>>>(
-        h, att_src, att_dst, n, n_dst, heads, chead, asrc, adst);
+    const int grid = grid_for(n * heads, BLOCK / DSUB);
+    if (h_bf16)
+        gat_dots_fwd_kernel<__hip_bfloat16><<<grid, BLOCK, 0, s>>>(
+            (const __hip_bfloat16*)h, att_src, att_dst, n, n_dst, heads,
+            chead, asrc, adst);
+    else
+        gat_dots_fwd_kernel<float><<<grid, BLOCK, 0, s>>>(
+            (const float*)h, att_src, att_dst, n, n_dst, heads, chead,
+            asrc, adst);
     QK_CHECK_HIP(hipGetLastError());
 }
 
-void launch_gat_dots_bwd(hipStream_t s, const float* h,
+void launch_gat_dots_bwd(hipStream_t s, const void* h, bool h_bf16,
                          const float* att_src, const float* att_dst,
                          const float* g_asrc, const float* g_adst,
                          int64_t n, int64_t n_dst, int heads, int chead,
-                         float* g_h, float* g_att_src, float* g_att_dst) {
+                         void* g_h, float* g_att_src, float* g_att_dst) {
     if (n == 0) return;
     if (chead % 4 != 0)
         throw std::runtime_error("gat_dots: chead must be a multiple of 4");
     const int D = heads * chead;
     if (D > 4096)
         throw std::runtime_error("gat_dots: heads*chead too large for LDS");
-    gat_dots_bwd_kernel<<<grid_for(n * heads, BLOCK / DSUB), BLOCK,
-                          2 * D * sizeof(float), s>>>(
-        h, att_src, att_dst, g_asrc, g_adst, n, n_dst, heads, chead, g_h,
-        g_att_src, g_att_dst);
+    const int grid = grid_for(n * heads, BLOCK / DSUB);
+    if (h_bf16)
+        gat_dots_bwd_kernel<__hip_bfloat16>
+            <<<grid, BLOCK, 2 * D * sizeof(float), s>>>(
+                (const __hip_bfloat16*)h, att_src, att_dst, g_asrc, g_adst,
+                n, n_dst, heads, chead, (__hip_bfloat16*)g_h, g_att_src,
+                g_att_dst);
+    else
+        gat_dots_bwd_kernel<float>
+            <<<grid, BLOCK, 2 * D * sizeof(float), s>>>(
+                (const float*)h, att_src, att_dst, g_asrc, g_adst, n,
+                n_dst, heads, chead, (float*)g_h, g_att_src, g_att_dst);
     QK_CHECK_HIP(hipGetLastError());
 }
 

@@ -308,21 +308,25 @@ class GATConv(nn.Module):
         n_dst = x_dst.size(0) if size is None else int(size[1])
 
         h2 = self.lin(x_src)
-        # bf16 models: projection GEMM and the feature path stay bf16 (the
-        # bandwidth win); attention math runs through the fp32 fused
-        # kernels via one boundary cast — torch's bf16 fallback chain
-        # (scatter_reduce amax -> ~120 rocprim sort kernels per step) is
-        # far slower than the cast
+        # bf16 models: the projected features stay bf16 end to end — the
+        # dots/wsum kernels take bf16 h with fp32 accumulation, while the
+        # LOGITS (alpha) and softmax stay fp32 (standard practice, and
+        # they are [*, H]-small).  Odd head widths fall back to one
+        # boundary cast; torch's bf16 softmax fallback chain
+        # (scatter_reduce amax -> ~120 rocprim sort kernels) is far
+        # slower than either.
         out_dtype = h2.dtype
-        if h2.is_cuda and h2.dtype == torch.bfloat16 and self.sorted_dst:
+        native = (h2.is_cuda and self.sorted_dst and C % 4 == 0
+                  and h2.dtype in (torch.float32, torch.bfloat16))
+        if (h2.is_cuda and h2.dtype == torch.bfloat16 and self.sorted_dst
+                and not native):
             h2 = h2.float()
         h_src = h2.view(-1, H, C)
         prefix = (x_dst.data_ptr() == x_src.data_ptr()
                   and x_dst.size(0) <= x_src.size(0)
                   and x_dst.stride() == x_src.stride()
                   and x_dst.size(1) == x_src.size(1))
-        if (prefix and h2.is_cuda and h2.dtype == torch.float32
-                and C % 4 == 0):
+        if prefix and native:
             # bipartite prefix convention (x_dst is x_src[:n_dst]) with
             # the fused dots kernel: one read of h for both logit sets
             alpha_src, alpha_dst = _GatDots.apply(
@@ -358,8 +362,9 @@ class GATConv(nn.Module):
         if self.training and self.dropout > 0:
             alpha = F.dropout(alpha, p=self.dropout)
 
-        if (self.sorted_dst and h_src.is_cuda
-                and h_src.dtype == torch.float32 and C % 4 == 0
+        if (self.sorted_dst and h_src.is_cuda and C % 4 == 0
+                and h_src.dtype in (torch.float32, torch.bfloat16)
+                and alpha.dtype == torch.float32
                 and dst.numel() > 0):
             # fused weighted segment sum over the dst-sorted edges
             if not fused:

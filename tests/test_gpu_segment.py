@@ -283,3 +283,51 @@ def test_gat_bf16_e2e_step():
         if p.grad is not None:
             assert p.grad.dtype == p.dtype, name
             assert torch.isfinite(p.grad.float()).all(), name
+
+
+def test_segment_wsum_bf16_matches_fp32():
+    """bf16 weighted segment sum (x bf16, weights fp32): fwd + both
+    backwards vs the fp32 reference."""
+    torch.manual_seed(0)
+    n_src, n_dst, H, C = 4000, 500, 4, 64
+    x32 = torch.randn(n_src, H * C, device="cuda")
+    x16 = x32.to(torch.bfloat16)
+    deg = torch.randint(1, 20, (n_dst,), device="cuda")
+    dst_ptr = torch.zeros(n_dst + 1, dtype=torch.long, device="cuda")
+    torch.cumsum(deg, 0, out=dst_ptr[1:])
+    E = int(dst_ptr[-1])
+    src = torch.randint(0, n_src, (E,), device="cuda")
+    w = torch.rand(E, H, device="cuda")
+    out16 = _ext.segment_wsum(x16, w, src, dst_ptr, H)
+    assert out16.dtype == torch.bfloat16
+    out32 = _ext.segment_wsum(x16.float(), w, src, dst_ptr, H)
+    assert torch.allclose(out16.float(), out32, atol=5e-2, rtol=5e-2)
+
+    g16 = torch.randn(n_dst, H * C, device="cuda").to(torch.bfloat16)
+    gx16, gw16 = _ext.segment_wsum_backward(g16, x16, w, src, dst_ptr, H,
+                                            True, True)
+    gx32, gw32 = _ext.segment_wsum_backward(g16.float(), x16.float(), w,
+                                            src, dst_ptr, H, True, True)
+    assert gx16.dtype == torch.bfloat16 and gw16.dtype == torch.float32
+    assert torch.allclose(gx16.float(), gx32, atol=8e-2, rtol=8e-2)
+    assert torch.allclose(gw16, gw32, atol=3e-2, rtol=3e-2)
+
+
+def test_gat_dots_bf16_matches_fp32():
+    from quiver.nn import _GatDots
+    torch.manual_seed(1)
+    N, H, C, n_dst = 4000, 4, 64, 500
+    h32 = torch.randn(N, H * C, device="cuda", requires_grad=True)
+    h16 = h32.detach().to(torch.bfloat16).requires_grad_(True)
+    a_s = torch.randn(H * C, device="cuda", requires_grad=True)
+    a_d = torch.randn(H * C, device="cuda", requires_grad=True)
+    s16, d16 = _GatDots.apply(h16, a_s, a_d, n_dst, H)
+    assert s16.dtype == torch.float32  # logits stay fp32
+    s32, d32 = _GatDots.apply(h16.detach().float().requires_grad_(False),
+                              a_s.detach(), a_d.detach(), n_dst, H)
+    assert torch.allclose(s16, s32, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(d16, d32, atol=1e-3, rtol=1e-3)
+    (s16.square().sum() + d16.square().sum()).backward()
+    assert h16.grad.dtype == torch.bfloat16
+    assert torch.isfinite(h16.grad.float()).all()
+    assert a_s.grad is not None and torch.isfinite(a_s.grad).all()
