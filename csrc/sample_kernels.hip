@@ -167,8 +167,16 @@ __global__ void cal_next_kernel(const int64_t* __restrict__ indptr,
 
 inline int grid_for(int64_t work_items, int per_block) {
     int64_t blocks = (work_items + per_block - 1) / per_block;
-    // 256 CUs x 8 blocks: cap and grid-stride the rest
-    if (blocks > 2048) blocks = 2048;
+    // 256 CUs x 8 blocks: cap and grid-stride the rest.
+    // QUIVER_SAMPLE_BLOCKS: experimental cap for UVA-mode tuning (the
+    // zero-copy CSR reads share the PCIe-contention physics of the
+    // feature gather).
+    static int env_cap = [] {
+        const char* e = getenv("QUIVER_SAMPLE_BLOCKS");
+        return e ? atoi(e) : 0;
+    }();
+    int64_t cap = env_cap > 0 ? env_cap : 2048;
+    if (blocks > cap) blocks = cap;
     if (blocks < 1) blocks = 1;
     return (int)blocks;
 }
