@@ -264,6 +264,8 @@ def main():
         opt.step()
         return sum(adj.edge_index.shape[1] for adj in adjs)
 
+    step_times = [] if os.environ.get("QUIVER_BENCH_STEP_TIMES") else None
+
     def run_range(lo, hi):
         total = 0
         if args.no_overlap:
@@ -271,6 +273,18 @@ def main():
                 n_id, bs, adjs = sampler.sample(batches[i])
                 x = feature[n_id]
                 total += train_on(n_id, bs, adjs, x)
+        elif step_times is not None:
+            pf = quiver.TrainingPrefetcher(sampler, feature,
+                                           batches[lo:hi], depth=2,
+                                           device=local_rank,
+                                           num_streams=args.prefetch_streams)
+            t_prev = time.perf_counter()
+            for n_id, bs, adjs, x in pf:
+                total += train_on(n_id, bs, adjs, x)
+                torch.cuda.synchronize()
+                now = time.perf_counter()
+                step_times.append((now - t_prev) * 1000)
+                t_prev = now
         else:
             pf = quiver.TrainingPrefetcher(sampler, feature,
                                            batches[lo:hi], depth=2,
@@ -317,6 +331,9 @@ def main():
         dist.all_reduce(e, op=dist.ReduceOp.SUM)
         edges_done = int(e)
 
+    if step_times is not None and rank == 0:
+        print("# step ms:", " ".join(f"{t:.2f}" for t in step_times),
+              flush=True)
     if rank == 0:
         ms_per_step = elapsed / args.steps * 1000
         steps_per_epoch = (n_train + args.batch * world - 1) // (args.batch *
