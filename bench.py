@@ -266,49 +266,36 @@ def main():
 
     step_times = [] if os.environ.get("QUIVER_BENCH_STEP_TIMES") else None
 
+    # ONE prefetcher spanning warmup + timed steps: a fresh prefetcher at
+    # the timing boundary would create a new side stream (fresh caching-
+    # allocator pool -> hundreds of MB re-allocated) and refill the
+    # pipeline inside the timed window — measured as a ~10 ms first timed
+    # step.  A continuous pipeline is also how a real epoch runs; the
+    # depth batches produced ahead of t0 cancel against the depth
+    # produced-but-unconsumed at the end.
+    pf_iter = None
+    if not args.no_overlap:
+        pf = quiver.TrainingPrefetcher(sampler, feature, batches, depth=2,
+                                       device=local_rank,
+                                       num_streams=args.prefetch_streams)
+        pf_iter = iter(pf)
+
     def run_range(lo, hi):
         total = 0
-        if args.no_overlap:
-            for i in range(lo, hi):
+        t_prev = time.perf_counter()
+        for i in range(lo, hi):
+            if pf_iter is None:
                 n_id, bs, adjs = sampler.sample(batches[i])
                 x = feature[n_id]
-                total += train_on(n_id, bs, adjs, x)
-        elif step_times is not None:
-            pf = quiver.TrainingPrefetcher(sampler, feature,
-                                           batches[lo:hi], depth=2,
-                                           device=local_rank,
-                                           num_streams=args.prefetch_streams)
-            t_prev = time.perf_counter()
-            for n_id, bs, adjs, x in pf:
-                total += train_on(n_id, bs, adjs, x)
+            else:
+                n_id, bs, adjs, x = next(pf_iter)
+            total += train_on(n_id, bs, adjs, x)
+            if step_times is not None:
                 torch.cuda.synchronize()
                 now = time.perf_counter()
                 step_times.append((now - t_prev) * 1000)
                 t_prev = now
-        else:
-            pf = quiver.TrainingPrefetcher(sampler, feature,
-                                           batches[lo:hi], depth=2,
-                                           device=local_rank,
-                                           num_streams=args.prefetch_streams)
-            for n_id, bs, adjs, x in pf:
-                total += train_on(n_id, bs, adjs, x)
         return total
-
-    if args.report_hit_rate:
-        # fraction of gathered rows served from the HBM hot cache
-        row_bytes = feat_dim * feat_cpu.element_size()
-        hot_rows = quiver.utils.parse_size(args.cache) // row_bytes
-        if args.cache_policy == "p2p_clique_replicate":
-            hot_rows *= world
-        hits = tot = 0
-        for i in range(args.warmup):
-            n_id, _, _ = sampler.sample(batches[i])
-            rows = feature.feature_order[n_id] \
-                if feature.feature_order is not None else n_id
-            hits += int((rows < hot_rows).sum())
-            tot += n_id.numel()
-        print(f"# hot-cache hit rate ({args.placement} placement): "
-              f"{hits/max(tot,1):.4f} over {tot} gathered rows", flush=True)
 
     run_range(0, args.warmup)
     if distributed:
