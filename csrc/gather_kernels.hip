@@ -84,9 +84,12 @@ inline int gather_max_blocks(const GatherSpec& spec, bool overlapped) {
     }();
     if (env_cap > 0) return env_cap;
     if (!spec.has_host_shard) return 2048;
-    // 112 = middle of the measured knee band (products best at 96,
-    // papers100M best at 128; both within ~0.1 ms of 112)
-    return overlapped ? 112 : 640;
+    // knee band measured per row size (ms/step sweeps on the bench):
+    // 400 B fp32 products rows: best 96-112; 200-256 B bf16 rows: best
+    // ~160 (smaller rows need more rows in flight for the same PCIe
+    // occupancy); papers 512 B rows within noise of 112-128.
+    if (overlapped) return spec.row_bytes <= 256 ? 160 : 112;
+    return 640;
 }
 
 template <typename VecT, bool SCATTER>
