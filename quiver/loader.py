@@ -26,7 +26,7 @@ __all__ = ["TrainingPrefetcher"]
 
 
 class _GraphedChain:
-    """`depth` alternating hipGraph captures of the sample->gather chain.
+    """depth+1 alternating hipGraph captures of the sample->gather chain.
 
     A captured graph replays the whole per-batch chain (~75 kernel
     launches) as ONE hipGraphLaunch, eliminating per-launch host cost.
@@ -38,9 +38,9 @@ class _GraphedChain:
 
     Each slot owns its own capture + static buffers: batch i and batch
     i+1 are in flight simultaneously, so a single capture would overwrite
-    live outputs.  The caller replays slot i%depth for batch i and must
-    order each replay after the consumer of that slot's previous batch
-    (an event on the main stream).
+    live outputs.  The caller replays slots round-robin and must order
+    each replay after the consumer of that slot's previous batch (an
+    event on the main stream; see produce()).
     """
 
     def __init__(self, sampler, feature, seeds_proto, stream, depth):
