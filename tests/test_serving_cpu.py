@@ -44,6 +44,7 @@ def test_cpu_serving_pipeline(serving_graph, tmp_path):
                                     sample_mode="CPU", csr_topo=topo,
                                     sizes=[4, 4], proc_num_per_device=1)
     server.start(join=False)
+    assert server.wait_ready(timeout=90) == 1  # warm-up barrier
 
     n_req = 5
     for i in range(n_req):
