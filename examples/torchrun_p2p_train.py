@@ -52,12 +52,8 @@ def main():
                              cache_policy="p2p_clique_replicate",
                              csr_topo=csr_topo)
 
-    def all_gather(obj):
-        objs = [None] * world
-        dist.all_gather_object(objs, obj)
-        return objs
-
-    feature.from_cpu_tensor_dist(feat_cpu, world, rank, all_gather)
+    # world/rank/all_gather default to the torch.distributed group
+    feature.from_cpu_tensor_dist(feat_cpu)
 
     model = GraphSAGE(dim, 128, classes, num_layers=2, dropout=0.0).to(device)
     model = torch.nn.parallel.DistributedDataParallel(

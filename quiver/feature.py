@@ -146,8 +146,8 @@ class Feature(object):
                 shard_tensor.append(self.cpu_part, -1)
                 self.clique_tensor_list[clique_id] = shard_tensor
 
-    def from_cpu_tensor_dist(self, cpu_tensor: torch.Tensor, world: int,
-                             rank: int, all_gather_object,
+    def from_cpu_tensor_dist(self, cpu_tensor: torch.Tensor, world=None,
+                             rank=None, all_gather_object=None,
                              score: torch.Tensor = None):
         """Collaborative p2p_clique_replicate build, one process per GPU.
 
@@ -165,9 +165,24 @@ class Feature(object):
 
         Reference design point: NVLink-sharded p2p store built by one
         owner + cudaIpc reopen (quiver_feature.cu:378-421, examples).
+
+        world/rank/all_gather_object default to the initialized
+        torch.distributed process group.
         """
         assert self.cache_policy == "p2p_clique_replicate", \
             "distributed build targets the sharded (p2p) layout"
+        if all_gather_object is None:
+            import torch.distributed as dist
+            assert dist.is_initialized(), \
+                "pass all_gather_object or init torch.distributed first"
+            world = world if world is not None else dist.get_world_size()
+            rank = rank if rank is not None else dist.get_rank()
+
+            def all_gather_object(obj, _w=world):
+                objs = [None] * _w
+                dist.all_gather_object(objs, obj)
+                return objs
+        assert world is not None and rank is not None
         budget = parse_size(self.device_cache_size)
         total_budget = budget * world
         cache_rows = min(self.cal_size(cpu_tensor, total_budget),
