@@ -72,7 +72,7 @@ def main():
         loss.backward()          # DDP allreduce over RCCL/xGMI
         opt.step()
         if rank == 0 and step % 5 == 0:
-            print(f"step {step}: loss {float(loss):.4f}", flush=True)
+            print(f"step {step}: loss {loss.detach().item():.4f}", flush=True)
     dist.barrier()
     if rank == 0:
         print("done")
