@@ -74,7 +74,7 @@ def _run_workers(target, args, world=WORLD):
 # from_cpu_tensor_dist: per-rank shard alloc + hipIpc reopen of peers
 # ---------------------------------------------------------------------------
 
-def _dist_feature_worker(rank, world, q, rdv_file):
+def _dist_feature_worker(rank, world, q, rdv_file, with_topo=True):
     try:
         dev = rank % torch.cuda.device_count()
         torch.cuda.set_device(dev)
@@ -83,7 +83,8 @@ def _dist_feature_worker(rank, world, q, rdv_file):
 
         indptr, indices, feat = _make_data()
         n, d = feat.shape
-        topo = quiver.CSRTopo(indptr=indptr, indices=indices)
+        topo = quiver.CSRTopo(indptr=indptr, indices=indices) \
+            if with_topo else None
         row_bytes = d * feat.element_size()
         # 600 of 1000 rows hot across the job -> per-rank budget 300 rows;
         # remainder exercises the pinned cold tier in the same store
@@ -134,7 +135,12 @@ def _dist_feature_worker(rank, world, q, rdv_file):
 
 
 def test_dist_p2p_feature_build(tmp_path):
-    _run_workers(_dist_feature_worker, (str(tmp_path / "rdv"),))
+    _run_workers(_dist_feature_worker, (str(tmp_path / "rdv"), True))
+
+
+def test_dist_p2p_feature_build_no_topo(tmp_path):
+    """No csr_topo: identity row order, direct slice path."""
+    _run_workers(_dist_feature_worker, (str(tmp_path / "rdv"), False))
 
 
 # ---------------------------------------------------------------------------
