@@ -143,6 +143,13 @@ def test_dist_p2p_feature_build_no_topo(tmp_path):
     _run_workers(_dist_feature_worker, (str(tmp_path / "rdv"), False))
 
 
+def test_dist_p2p_feature_build_world3(tmp_path):
+    """3 ranks: each opens TWO peers' hipIpc shards (closer to the
+    8-rank SCALE shape than the pairwise case)."""
+    _run_workers(_dist_feature_worker, (str(tmp_path / "rdv"), True),
+                 world=3)
+
+
 # ---------------------------------------------------------------------------
 # native RCCL communicator with one process per rank
 # ---------------------------------------------------------------------------
