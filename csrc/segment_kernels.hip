@@ -141,9 +141,27 @@ segment_mean_fwd_bf16_kernel(const __hip_bfloat16* __restrict__ x,
         for (int64_t c = (int64_t)lane * VPL_BF; c < dim; c += SUB * VPL_BF) {
             float acc[VPL_BF] = {};
             const int w = (int)min((int64_t)VPL_BF, dim - c);
+            // dim % 4 == 0 makes every row base 8-byte aligned: use two
+            // bf16x4 loads per 8-channel chunk instead of four bf16x2
+            const bool v8 = (dim & 3) == 0;
             for (int64_t e = beg; e < end; ++e) {
                 const __hip_bfloat16* row = x + src[e] * dim + c;
-                if (w == VPL_BF) {
+                if (w == VPL_BF && v8) {
+#pragma unroll
+                    for (int i = 0; i < 2; ++i) {
+                        struct alignas(8) bf16x4v {
+                            __hip_bfloat162 lo, hi;
+                        };
+                        const bf16x4v v4 =
+                            *reinterpret_cast<const bf16x4v*>(row + 4 * i);
+                        const float2 f0 = __bfloat1622float2(v4.lo);
+                        const float2 f1 = __bfloat1622float2(v4.hi);
+                        acc[4 * i] += f0.x;
+                        acc[4 * i + 1] += f0.y;
+                        acc[4 * i + 2] += f1.x;
+                        acc[4 * i + 3] += f1.y;
+                    }
+                } else if (w == VPL_BF) {
 #pragma unroll
                     for (int i = 0; i < 4; ++i) {
                         const float2 f = __bfloat1622float2(
