@@ -215,6 +215,10 @@ class Feature(object):
             own.shard_tensor.append(rows(beg, end).contiguous(), self.rank)
             handles = [it.share_ipc()
                        for it in own.shard_tensor.share_ipc()]
+        # keep this rank's allocation alive on SELF before any fallible
+        # step past the handle exchange: peers map it via hipIpc, and a
+        # failure-then-GC here would free memory they hold open
+        self._dist_own_keepalive = own
         gathered = all_gather_object((rank, self.rank, handles))
 
         # xGMI peer access across every device that owns a shard
