@@ -113,3 +113,44 @@ def test_auto_despatch_routing(tmp_path):
     assert t_cpu > 0 and t_gpu > 0  # arrival stamps ride with requests
     assert isinstance(gpu_q.get(timeout=10), _Stop)
     assert isinstance(cpu_q.get(timeout=10), _Stop)
+
+
+def test_drain_after_stop_single_sentinel(tmp_path):
+    """_drain_after_stop consumes sentinels while draining and re-queues
+    exactly ONE on exit (round-1 advisor finding: sentinels accumulated)."""
+    import queue as pyqueue
+    from quiver.serving import _drain_after_stop
+    q = pyqueue.Queue()
+    q.put("a")
+    q.put(_Stop())
+    q.put("b")
+    q.put(_Stop())
+    stop = _Stop()
+    got = list(_drain_after_stop(q, stop))
+    assert got == ["a", "b"]
+    leftovers = []
+    while not q.empty():
+        leftovers.append(q.get_nowait())
+    assert len(leftovers) == 1 and isinstance(leftovers[0], _Stop)
+
+
+def test_auto_despatch_preparation_mode(tmp_path):
+    """Preparation mode fans every request to BOTH queues (warm-up path),
+    with arrival timestamps attached."""
+    import numpy as np
+    nn = np.ones(100, dtype=np.int64)
+    path = tmp_path / "nn.npy"
+    np.save(path, nn)
+    batcher = quiver.RequestBatcher(
+        device_num=1, stream_queue_list=[mp.get_context("spawn").Queue()],
+        input_proc_per_device=0, sample_mode="Auto",
+        request_mode="Preparation", threshold=10, neighbour_path=str(path))
+    sq = batcher.stream_queue_list[0]
+    sq.put(np.array([1, 2]))
+    sq.put(_Stop())
+    batcher.auto_despatch(0)
+    for q in (batcher.cpu_batched_queue_list[0],
+              batcher.gpu_batched_queue_list[0]):
+        ids, t = q.get(timeout=10)
+        assert list(ids) == [1, 2] and t > 0
+        assert isinstance(q.get(timeout=10), _Stop)
