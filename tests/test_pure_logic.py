@@ -71,3 +71,37 @@ def test_parse_size_units():
     assert parse_size("2G") == 2 * 2**30
     assert parse_size(12345) == 12345
     assert parse_size("8") == 8
+
+
+def test_partition_info_dispatch_cpu():
+    """PartitionInfo: host bucketing + replicated hot nodes folded into
+    the local range (reference feature.py:461-527), pure tensor logic."""
+    from quiver.feature import PartitionInfo
+    g2h = torch.tensor([0, 0, 1, 1, 2, 2, 0, 1])   # 8 nodes on 3 hosts
+    replicate = torch.tensor([2, 4])                # hot nodes mirrored here
+    info = PartitionInfo(device="cpu", host=0, hosts=3,
+                         global2host=g2h.clone(), replicate=replicate)
+    # replicated remote nodes now claim host 0
+    assert int(info.global2host[2]) == 0 and int(info.global2host[4]) == 0
+    ids = torch.tensor([0, 2, 3, 4, 5, 7])
+    host_ids, host_orders = info.dispatch(ids)
+    assert len(host_ids) == 3
+    # host 0 serves: 0 (native), 2 and 4 (replicas) -> positions 0,1,3
+    assert host_orders[0].tolist() == [0, 1, 3]
+    # host 1 serves 3 and 7; host 2 serves 5
+    assert host_orders[1].tolist() == [2, 5]
+    assert host_orders[2].tolist() == [4]
+    # every queried id lands on exactly one host
+    total = sum(o.numel() for o in host_orders)
+    assert total == ids.numel()
+
+
+def test_topo_single_clique_without_gpu():
+    """GPU-less Topo: all-true access matrix -> one clique, stable ids."""
+    from quiver.utils import Topo
+    t = Topo([0, 1, 2, 3])
+    assert t.p2p_clique_count == 1
+    assert t.cliques == [[0, 1, 2, 3]]
+    for d in range(4):
+        assert t.get_clique_id(d) == 0
+    assert t.p2p_clique(2) == [0, 1, 2, 3]
